@@ -71,10 +71,16 @@ def best_nf_grid(n_max: int, freq_levels, pc: PowerCoeffs, tc: LatencyCoeffs,
 def freq_for_perf_expand(n0: int, f0: float, n1: int, tc: LatencyCoeffs,
                          freq_levels: Iterable[float]) -> float:
     """Pick the ladder frequency that preserves T(n0, f0) after growing to n1
-    GPUs (reference keep_perf_when_expand, policy_paper.py:19-29 — dead code
-    there; exposed here as a usable utility)."""
+    GPUs.  Capability parity with the reference keep_perf_when_expand
+    (policy_paper.py:19-29 — dead code there), but with the T(n, f) inversion
+    done correctly for n1 > 1: beta/f = T*n - alpha - gamma*n (the reference's
+    formula omits the *n on T, valid only at n1 == 1)."""
+    n1 = max(1, int(n1))
     T_target = unit_time_s(n0, max(1e-9, f0), tc)
-    denom = T_target - tc.alpha_t - tc.gamma_t * max(1, int(n1))
+    if n1 == 1:
+        denom = T_target - tc.alpha_t
+    else:
+        denom = T_target * n1 - tc.alpha_t - tc.gamma_t * n1
     if denom <= 1e-12:
         return f0
     f_cont = tc.beta_t / denom
