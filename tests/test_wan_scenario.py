@@ -46,9 +46,12 @@ def test_paper_scenario_shape(paper_sc):
     i = sc.ingress_names.index("gw-us-west")
     d = sc.dc_names.index("us-west")
     assert sc.wan_latency_s[i][d] == pytest.approx(0.012)
-    # multihop: gw-us-west -> eu-west has no direct edge; via us-east (70+90)
+    # multihop: gw-us-west -> eu-west has no direct edge; shortest is via
+    # eu-central (110) -> gw-eu-west (20) -> eu-west (10) = 140 ms
     d2 = sc.dc_names.index("eu-west")
-    assert sc.wan_latency_s[i][d2] == pytest.approx(0.160)
+    assert sc.wan_latency_s[i][d2] == pytest.approx(0.140)
+    assert sc.wan_paths[("gw-us-west", "eu-west")] == \
+        ["gw-us-west", "eu-central", "gw-eu-west", "eu-west"]
     # carbon only for 3 DCs
     cv = sc.carbon_vec()
     assert (cv > 0).sum() == 3
