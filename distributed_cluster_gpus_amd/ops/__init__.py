@@ -1,0 +1,46 @@
+"""Native extension loaders.
+
+``load_des_core()`` — the C++ scalar DES core (CPU; always expected to build).
+``load_sim_hip()``  — the batched MI355X HIP engine.  On a machine with a GPU
+this must NOT silently fall back: engines fail loudly if the extension is
+missing so GPU tests can never pass on an eager-Python substitute.
+"""
+import importlib
+import os
+
+
+def load_des_core():
+    try:
+        return importlib.import_module("distributed_cluster_gpus_amd.ops._des_core")
+    except ImportError as e:
+        raise ImportError(
+            "_des_core native extension not built. Run "
+            "`python setup.py build_ext --inplace` at the repo root."
+        ) from e
+
+
+def have_des_core() -> bool:
+    try:
+        load_des_core()
+        return True
+    except ImportError:
+        return False
+
+
+def load_sim_hip():
+    try:
+        return importlib.import_module("distributed_cluster_gpus_amd.ops._sim_hip")
+    except ImportError as e:
+        raise ImportError(
+            "_sim_hip (gfx950 HIP engine) not built. Run "
+            "`python -m distributed_cluster_gpus_amd.ops.build_hip` "
+            "(requires hipcc; cross-compiles fine without a GPU)."
+        ) from e
+
+
+def have_sim_hip() -> bool:
+    try:
+        load_sim_hip()
+        return True
+    except ImportError:
+        return False
