@@ -1,0 +1,185 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: batched Monte-Carlo replica engine on the paper
+multi-DC config (BASELINE.json metric: simulated events/sec, whole node).
+
+One step = every replica advancing by `--events-per-step` simulation events
+through the gfx950 advance kernel.  Weak scaling: each GPU owns
+`--replicas-per-gpu` replicas, so per-GPU work is fixed as N grows; the
+reported value is the WHOLE-JOB aggregate events/sec over all ranks.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  # N>1: torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N ...
+
+Reference baseline: 4,003 events/sec — the pure-Python
+reference simulator, default_policy, paper topology, sinusoid inference
+arrivals 6/s amp 0.6 period 300 + poisson training 0.3/s, measured on one
+CPU core.  Same algorithm, same topology, same arrival processes here.
+
+A secondary RL metric (CHSAC-AF SAC updates/sec on-device, batch 256 —
+BASELINE.md: 36.9/s on CPU) is measured OUTSIDE the timed region and
+reported inside config.
+"""
+import argparse
+import json
+import os
+import time
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--replicas-per-gpu", type=int, default=4096)
+    p.add_argument("--events-per-step", type=int, default=1000)
+    p.add_argument("--algo", type=str, default="default_policy")
+    p.add_argument("--seed", type=int, default=123)
+    p.add_argument("--with-rl", type=int, default=1,
+                   help="also measure CHSAC-AF SAC updates/sec (untimed region)")
+    return p.parse_args()
+
+
+def measure_rl_steps_per_sec(device, n_updates=60, batch=256):
+    """CHSAC-AF train-step rate on the paper obs space (49-dim, 8 DC, 8 g)."""
+    import torch
+
+    from distributed_cluster_gpus_amd.rl.agent import CHSACAgent, CHSACAgentConfig
+    from distributed_cluster_gpus_amd.rl.replay import ReplayRing
+    torch.manual_seed(0)
+    obs_dim = 49
+    agent = CHSACAgent(CHSACAgentConfig(
+        obs_dim=obs_dim, n_dc=8, n_g_choices=8,
+        constraints={"latency_p99": 500.0, "gpu_over": 0.0}, device=str(device)))
+    ring = ReplayRing(capacity=4096, obs_dim=obs_dim, n_costs=2,
+                      cost_names=["latency_p99", "gpu_over"], n_dc=8, n_g=8,
+                      device=str(device), seed=0)
+    B = 4096
+    ring.add_batch(s=torch.randn(B, obs_dim), s_next=torch.randn(B, obs_dim),
+                   a_dc=torch.randint(0, 8, (B,)), a_g=torch.randint(0, 8, (B,)),
+                   r=torch.randn(B), costs=torch.rand(B, 2) * 100,
+                   done=torch.ones(B))
+    for _ in range(10):  # warmup
+        agent.train_step(ring.sample(batch))
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(n_updates):
+        agent.train_step(ring.sample(batch))
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    return n_updates / (time.perf_counter() - t0)
+
+
+def main():
+    args = parse_args()
+    import torch
+
+    from distributed_cluster_gpus_amd.configs.paper import (build_arrivals,
+                                                            paper_scenario)
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.parallel.dist import (allreduce_scalar,
+                                                            barrier,
+                                                            init_distributed,
+                                                            is_distributed)
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        init_distributed("nccl")
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if not torch.cuda.is_available():
+        raise RuntimeError("bench.py requires a ROCm GPU")
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    total_replicas = args.replicas_per_gpu * world
+    total_steps = args.warmup + args.steps
+    # duration generous enough that no replica reaches end_time mid-bench
+    # (~150-200 events per simulated second per replica on this workload)
+    duration = max(1200.0, total_steps * args.events_per_step / 100.0)
+    qcap = int(max(2048, 0.4 * duration))
+
+    sc = paper_scenario()
+    inf, trn = build_arrivals()  # sinusoid 6/s amp .6 period 300; poisson 0.3/s
+    eng = BatchedEngine(sc, inf, trn, algo=args.algo,
+                        replicas=total_replicas, duration=duration,
+                        log_interval=5.0, out_dir=None, seed=args.seed,
+                        device=device, rank=rank, world=world, qcap=qcap,
+                        enable_logs=False)
+
+    ev = eng.t["ev_count"]
+
+    def step():
+        eng._sim.advance(duration, args.events_per_step)
+
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize(device)
+    ev_start = int(ev.sum().item())
+
+    barrier()
+    torch.cuda.synchronize(device)
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize(device)
+    barrier()
+    t1 = time.perf_counter()
+
+    if int(eng.t["err"].max().item()) != 0:
+        raise RuntimeError("engine error flags set during bench")
+    if int(eng.t["done"].max().item()) != 0:
+        raise RuntimeError("a replica finished inside the timed region; "
+                           "raise --events-per-step headroom (invalid run)")
+
+    elapsed = t1 - t0
+    events = float(int(ev.sum().item()) - ev_start)
+    # whole-job aggregate: sum events over ranks; MAX elapsed over ranks
+    if is_distributed():
+        events = allreduce_scalar(events, device=device)
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    value = events / elapsed
+
+    rl_steps_s = None
+    if args.with_rl and rank == 0:
+        rl_steps_s = measure_rl_steps_per_sec(device)
+
+    if rank == 0:
+        baseline = 4003.0  # BASELINE.md reference events/sec (CPU, 1 core)
+        out = {
+            "metric": "sim_events_per_sec",
+            "value": value,
+            "unit": "events/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": value / baseline,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "paper_config multi-DC geo-cluster DES",
+                "global_batch": total_replicas,
+                "seq_len": args.events_per_step,
+                "parallelism": f"replica_shard_dp{world}",
+                "algo": args.algo,
+                "replicas_per_gpu": args.replicas_per_gpu,
+                "events_per_step": args.events_per_step,
+                "arrivals": "sinusoid inf 6/s amp 0.6 period 300 + poisson trn 0.3/s",
+                "topology": "8 DC / 1488 GPUs / 8 ingresses",
+                "rl_train_steps_per_sec": rl_steps_s,
+                "rl_baseline_steps_per_sec": 36.9,
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

@@ -28,14 +28,24 @@ def have_des_core() -> bool:
 
 
 def load_sim_hip():
-    try:
-        return importlib.import_module("distributed_cluster_gpus_amd.ops._sim_hip")
-    except ImportError as e:
+    """Import the gfx950 batched-engine extension.  torch must be imported
+    first (the .so links against libtorch)."""
+    import importlib.util
+    import torch  # noqa: F401  (symbol provider)
+    path = os.path.join(os.path.dirname(__file__), "_sim_hip.so")
+    if not os.path.exists(path):
         raise ImportError(
-            "_sim_hip (gfx950 HIP engine) not built. Run "
+            "_sim_hip.so (gfx950 HIP engine) not built. Run "
             "`python -m distributed_cluster_gpus_amd.ops.build_hip` "
-            "(requires hipcc; cross-compiles fine without a GPU)."
-        ) from e
+            "(requires hipcc; cross-compiles fine without a GPU).")
+    import sys
+    if "_sim_hip" in sys.modules:
+        return sys.modules["_sim_hip"]
+    spec = importlib.util.spec_from_file_location("_sim_hip", path)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    sys.modules["_sim_hip"] = mod
+    return mod
 
 
 def have_sim_hip() -> bool:

@@ -1,0 +1,945 @@
+// Batched Monte-Carlo replica engine for MI355X (gfx950, CDNA4).
+//
+// Architecture (SURVEY §7 "fixed-Δt lockstep engine with exact event times"):
+// each CDNA4 wavefront (64 lanes) owns ONE replica of the multi-DC simulation
+// and advances it event-by-event with exact event times; thousands of
+// replicas run concurrently (256 CUs x 4 SIMDs x N waves).  All per-replica
+// state lives in HBM3E as replica-major structure-of-arrays, so lane-strided
+// scans (job-slot min-finish, empty-slot search, queue walks) are coalesced.
+// Next-event selection is a wave argmin over {16 arrival streams, in-flight
+// WAN transfers, per-DC cached min finish times, next log tick}.  The
+// (n, f) grid search maps the full 8x8 candidate grid onto the 64 lanes of
+// one wavefront (sim_models.hpp::wave_grid_argmin).
+//
+// Execution discipline: control flow is wave-uniform — every lane computes
+// the same scalar values redundantly; global-memory writes are guarded to
+// lane 0; cooperative phases (scans/argmins) use lane-strided reads + shfl
+// reductions.  Replicas are independent, so kernels need no inter-workgroup
+// communication; multi-GPU scaling shards replicas per rank
+// (parallel/sharding.py) with RCCL used only for metric reductions.
+//
+// Event semantics mirror the scalar engines (engine/oracle.py ==
+// reference simulator simcore/simulator_paper_multi.py:412-480): per-event
+// energy/util accrual using per-job f_used power, queue-drain on completion
+// with inference priority, per-algorithm admission decisions, the
+// finish%log_interval job-unit remainder quirk, the end-of-run baseline-model
+// energy flush.  RNG is Philox (philox.hpp) — parity with the scalar engines
+// is distributional, not bitwise (SURVEY §7 "RNG stream equivalence").
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <vector>
+
+#include "philox.hpp"
+#include "sim_models.hpp"
+
+namespace dcg {
+
+constexpr int MAX_DC = 8;
+constexpr int MAX_FREQ = 8;
+
+enum Algo { A_DEFAULT = 0, A_CAP_UNIFORM, A_CAP_GREEDY, A_JOINT_NF, A_BANDIT,
+            A_CARBON_COST, A_ECO_ROUTE, A_DEBUG };
+
+enum Err : int { ERR_NONE = 0, ERR_QUEUE_OVF = 1, ERR_XFER_OVF = 2,
+                 ERR_SLOT_OVF = 4, ERR_LOG_OVF = 8 };
+
+struct EngineDesc {
+  // sizes
+  int n_rep, n_dc, n_ing, n_freq, n_streams, total_slots, tcap, qcap;
+  double end_time, log_interval;
+  // scenario constants (device)
+  const double* freq_levels;      // [n_freq]
+  const double* pc;               // [dc][2][3]
+  const double* lc;               // [dc][2][3]
+  const double* wan_lat;          // [ing][dc]
+  const double* wan_bw;           // [ing][dc]
+  const double* carbon;           // [dc]
+  const double* price24;          // [24]
+  const int* total_gpus;          // [dc]
+  const double* p_idle;           // [dc]
+  const double* p_sleep;          // [dc]
+  const double* p_peak;           // [dc] (baseline end-flush model)
+  const double* pow_alpha;        // [dc]
+  const int* power_gating;        // [dc]
+  const double* default_freq;     // [dc]
+  const int* slot_off;            // [dc+1]
+  const int* slot_dc;             // [total_slots] slot -> dc
+  // policy / run params
+  int algo, max_gpj, inf_priority, scale_out_low, energy_aware;
+  double dvfs_low, dvfs_high, power_cap;
+  int eco_obj, num_fixed;
+  double fixed_freq;
+  double payload_gb[2];
+  // arrival processes by jtype: mode 0=poisson 1=sinusoid 2=off
+  int arr_mode[2];
+  double arr_rate[2], arr_amp[2], arr_period[2];
+  uint64_t seed;
+  int64_t rep_id_offset;          // global replica id of local replica 0
+  // per-replica scalars
+  double* now;                    // [r] (last event time; <0 = no event yet)
+  double* next_log;               // [r]
+  uint64_t* rng_ctr;              // [r]
+  int* jid_ctr;                   // [r]
+  int* done;                      // [r]
+  int* err;                       // [r]
+  double* arr_next;               // [r][n_streams]  (stream = ing*2 + jtype)
+  // per (r, dc)
+  int* busy;
+  float* cur_freq;
+  double* energy_j;
+  double* util_time;
+  double* util_begin;             // [r][dc] first-event stamp (<0 unset)
+  double* acc_unit;
+  double* p_active;               // cached sum of running-job powers
+  double* sum_tpt;                // cached sum of running-job throughputs
+  int* n_running;
+  double* dc_min_finish;          // cached min finish time (INF if none)
+  int* dc_min_slot;
+  // job slots [r][total_slots]
+  double* s_finish;               // INF = empty
+  double* s_start;
+  float* s_size;
+  float* s_fused;
+  float* s_netlat;
+  int* s_jid;
+  short* s_gpus;                  // 0 = empty
+  char* s_jtype;
+  char* s_ing;
+  // in-flight transfers [r][tcap]
+  double* x_time;                 // INF = empty
+  float* x_size;
+  float* x_netlat;
+  int* x_jid;
+  char* x_dc;
+  char* x_jtype;
+  char* x_ing;
+  // queues [r][dc][2] ring of capacity qcap
+  int* q_head;
+  int* q_len;
+  float* q_size;                  // [r][dc][2][qcap]
+  float* q_netlat;
+  int* q_jid;
+  char* q_ing;
+  // bandit state [r][dc][2][n_freq]
+  int* b_n;
+  float* b_s;
+  long long* b_t;                 // [r]
+  // metrics [r]
+  long long* ev_count;
+  long long* jobs_done;
+  long long* jobs_done_inf;
+  double* sum_lat;
+  double* sum_lat_inf;
+  double* sum_wait;               // queueing delay (start - dc arrival)
+  // logging (one designated replica)
+  int log_replica;                // -1 = off
+  int cl_cap, jl_cap;
+  int* cl_count;                  // [1]
+  double* cl_rows;                // [cl_cap][16]
+  int* jl_count;                  // [1]
+  double* jl_rows;                // [jl_cap][10]
+};
+
+// ---------------- wave helpers ----------------
+__device__ __forceinline__ int wave_sum_i32(int v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+__device__ __forceinline__ void store_fence() {
+  // make lane-0 global stores visible to this wave's subsequent loads
+  __builtin_amdgcn_s_waitcnt(0);
+}
+
+// ---------------- replica context ----------------
+struct Ctx {
+  const EngineDesc* S;
+  int r;        // local replica index
+  int lane;
+  PhiloxState rng;
+  double now;       // last processed event time (or -1)
+  double t_first;   // first event time (-1 until known)
+
+  __device__ const double* pc3(int d, int jt) const { return &S->pc[(d * 2 + jt) * 3]; }
+  __device__ const double* lc3(int d, int jt) const { return &S->lc[(d * 2 + jt) * 3]; }
+  __device__ int free_gpus(int d) const {
+    return S->total_gpus[d] - S->busy[r * S->n_dc + d];
+  }
+  __device__ double dc_power(int d) const {
+    int idle = free_gpus(d);
+    double pi = S->power_gating[d] ? S->p_sleep[d] : S->p_idle[d];
+    return S->p_active[r * S->n_dc + d] + idle * pi;
+  }
+  __device__ double price_kwh(double t) const {
+    int h = static_cast<int>(fmod(t, 86400.0) / 3600.0);
+    return S->price24[h];
+  }
+};
+
+// recompute a DC's cached min finish (wave-cooperative)
+__device__ void rescan_dc_min(Ctx& c, int d) {
+  const EngineDesc& S = *c.S;
+  int lo = S.slot_off[d], hi = S.slot_off[d + 1];
+  double v = D_INF;
+  int slot = -1;
+  for (int k = lo + c.lane; k < hi; k += 64) {
+    double f = S.s_finish[(int64_t)c.r * S.total_slots + k];
+    if (f < v) { v = f; slot = k; }
+  }
+  int wl;
+  double best = wave_argmin_f64(v, wl);
+  slot = __shfl(slot, wl, 64);
+  if (c.lane == 0) {
+    S.dc_min_finish[c.r * S.n_dc + d] = best;
+    S.dc_min_slot[c.r * S.n_dc + d] = best < D_INF ? slot : -1;
+  }
+  store_fence();
+}
+
+// start a job on DC d with (n, f); assumes free >= 1; wave-cooperative.
+__device__ void start_job(Ctx& c, int d, int jt, float size, float netlat,
+                          int jid, int ing, int n, double f, double now) {
+  const EngineDesc& S = *c.S;
+  int64_t base = (int64_t)c.r * S.total_slots;
+  int lo = S.slot_off[d], hi = S.slot_off[d + 1];
+  // find first empty slot (s_gpus == 0)
+  int cand = INT_MAX;
+  for (int k = lo + c.lane; k < hi; k += 64) {
+    if (S.s_gpus[base + k] == 0) { cand = k; break; }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    cand = min(cand, __shfl_xor(cand, off, 64));
+  if (cand == INT_MAX) {  // cannot happen if capacities == total_gpus
+    if (c.lane == 0) atomicOr(&S.err[c.r], ERR_SLOT_OVF);
+    return;
+  }
+  double T = d_unit_time(n, f, c.lc3(d, jt));
+  double finish = now + (double)size * T;
+  if (c.lane == 0) {
+    S.s_finish[base + cand] = finish;
+    S.s_start[base + cand] = now;
+    S.s_size[base + cand] = size;
+    S.s_fused[base + cand] = (float)f;
+    S.s_netlat[base + cand] = netlat;
+    S.s_jid[base + cand] = jid;
+    S.s_gpus[base + cand] = (short)n;
+    S.s_jtype[base + cand] = (char)jt;
+    S.s_ing[base + cand] = (char)ing;
+    int rd = c.r * S.n_dc + d;
+    S.busy[rd] += n;
+    S.n_running[rd] += 1;
+    S.p_active[rd] += d_job_power(n, f, c.pc3(d, jt));
+    S.sum_tpt[rd] += 1.0 / T;
+    if (finish < S.dc_min_finish[rd]) {
+      S.dc_min_finish[rd] = finish;
+      S.dc_min_slot[rd] = cand;
+    }
+  }
+  store_fence();
+}
+
+// heuristic allocator (policy.py:16-41 semantics); mutates cur_freq; returns g
+__device__ int heuristic_alloc(Ctx& c, int d, int jt) {
+  const EngineDesc& S = *c.S;
+  int rd = c.r * S.n_dc + d;
+  int free = c.free_gpus(d);
+  int g = free > 0 ? min(free, S.max_gpj) : 0;
+  double cf = S.cur_freq[rd];
+  double nf = cf;
+  if (!S.energy_aware) {  // perf_first
+    if (jt == 0) nf = S.dvfs_high;
+    else {
+      int qi = S.q_len[(rd) * 2 + 0];
+      nf = fmax(cf, qi > 0 ? S.dvfs_high : S.default_freq[d]);
+    }
+  } else {
+    if (jt == 0) nf = S.dvfs_high;
+    else if (S.scale_out_low && free >= 2) {
+      nf = S.dvfs_low;
+      g = min(free, S.max_gpj);
+    } else nf = fmax(cf, S.dvfs_low);
+  }
+  if (c.lane == 0) S.cur_freq[rd] = (float)nf;
+  store_fence();
+  return max(1, g);
+}
+
+// per-algorithm (n, f) decision at admission; returns chosen n, f
+template <int ALGO>
+__device__ void decide_nf(Ctx& c, int d, int jt, float size, double now,
+                          int& n_out, double& f_out) {
+  const EngineDesc& S = *c.S;
+  int free = c.free_gpus(d);
+  if (ALGO == A_JOINT_NF) {
+    GridPick g = wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                                  S.n_freq, S.max_gpj, 0, 0.0, 0.0, false, 0.0);
+    n_out = g.n; f_out = g.f;
+  } else if (ALGO == A_BANDIT) {
+    // UCB1 (learners.py:20-36): uniform-redundant serial loop over arms
+    long long t = S.b_t[c.r] + 1;
+    if (c.lane == 0) S.b_t[c.r] = t;
+    int64_t ab = ((int64_t)c.r * S.n_dc + d) * 2 + jt;
+    double bf = -1.0;
+    for (int k = 0; k < S.n_freq; ++k)
+      if (S.b_n[ab * S.n_freq + k] < 1) { bf = S.freq_levels[k]; break; }
+    if (bf < 0) {
+      double best_ucb = -1e9;
+      for (int k = 0; k < S.n_freq; ++k) {
+        int nn = S.b_n[ab * S.n_freq + k];
+        double mean = S.b_s[ab * S.n_freq + k] / nn;
+        double ucb = mean + sqrt(2.0 * log((double)t) / nn);
+        if (ucb > best_ucb) { best_ucb = ucb; bf = S.freq_levels[k]; }
+      }
+    }
+    n_out = min(free, S.max_gpj);
+    f_out = bf;
+  } else if (ALGO == A_CARBON_COST) {
+    double price = c.price_kwh(now);
+    GridPick g = (price > 0.0)
+        ? wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                           S.n_freq, S.max_gpj, 2, 0.0, price, false, 0.0)
+        : wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                           S.n_freq, S.max_gpj, 1, S.carbon[d], 0.0, false, 0.0);
+    n_out = g.n; f_out = g.f;
+  } else if (ALGO == A_DEBUG) {
+    n_out = S.num_fixed;
+    f_out = S.fixed_freq > 0 ? S.fixed_freq
+            : wave_energy_freq(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                               S.n_freq, S.num_fixed);
+  } else {  // heuristic family
+    n_out = heuristic_alloc(c, d, jt);
+    f_out = S.cur_freq[c.r * S.n_dc + d];
+  }
+}
+
+// queue ops (wave-uniform; lane-0 writes)
+__device__ bool queue_push(Ctx& c, int d, int jt, float size, float netlat,
+                           int jid, int ing) {
+  const EngineDesc& S = *c.S;
+  int q = (c.r * S.n_dc + d) * 2 + jt;
+  int len = S.q_len[q];
+  if (len >= S.qcap) {
+    if (c.lane == 0) atomicOr(&S.err[c.r], ERR_QUEUE_OVF);
+    return false;
+  }
+  int pos = (S.q_head[q] + len) % S.qcap;
+  if (c.lane == 0) {
+    int64_t at = (int64_t)q * S.qcap + pos;
+    S.q_size[at] = size;
+    S.q_netlat[at] = netlat;
+    S.q_jid[at] = jid;
+    S.q_ing[at] = (char)ing;
+    S.q_len[q] = len + 1;
+  }
+  store_fence();
+  return true;
+}
+
+__device__ bool queue_pop(Ctx& c, int d, int jt, float& size, float& netlat,
+                          int& jid, int& ing) {
+  const EngineDesc& S = *c.S;
+  int q = (c.r * S.n_dc + d) * 2 + jt;
+  int len = S.q_len[q];
+  if (len <= 0) return false;
+  int pos = S.q_head[q];
+  int64_t at = (int64_t)q * S.qcap + pos;
+  size = S.q_size[at];
+  netlat = S.q_netlat[at];
+  jid = S.q_jid[at];
+  ing = S.q_ing[at];
+  if (c.lane == 0) {
+    S.q_head[q] = (pos + 1) % S.qcap;
+    S.q_len[q] = len - 1;
+  }
+  store_fence();
+  return true;
+}
+
+// drain queues after a completion (inference first; reference :839-927)
+template <int ALGO>
+__device__ void drain_queues(Ctx& c, int d, double now) {
+  const EngineDesc& S = *c.S;
+  while (c.free_gpus(d) > 0) {
+    float size, netlat;
+    int jid, ing;
+    int jt;
+    if (S.inf_priority && queue_pop(c, d, 0, size, netlat, jid, ing)) jt = 0;
+    else if (queue_pop(c, d, 1, size, netlat, jid, ing)) jt = 1;
+    else break;
+    int n; double f;
+    decide_nf<ALGO>(c, d, jt, size, now, n, f);
+    n = max(1, min(n, c.free_gpus(d)));
+    start_job(c, d, jt, size, netlat, jid, ing, n, f, now);
+  }
+}
+
+// accrue energy + util across DCs to time t (lanes 0..n_dc-1)
+__device__ void accrue_to(Ctx& c, double t) {
+  const EngineDesc& S = *c.S;
+  if (c.lane < S.n_dc) {
+    int rd = c.r * S.n_dc + c.lane;
+    double last = c.now;
+    if (last < 0.0) {
+      S.util_begin[rd] = t;
+    } else {
+      double dt = fmax(0.0, t - last);
+      S.util_time[rd] += S.busy[rd] * dt;
+      S.energy_j[rd] += c.dc_power(c.lane) * dt;
+    }
+  }
+  store_fence();
+}
+
+// emit one cluster-log row set (logging replica only; wave-cooperative counts)
+__device__ void emit_cluster_rows(Ctx& c, double now) {
+  const EngineDesc& S = *c.S;
+  if (c.r != S.log_replica) return;
+  int64_t base = (int64_t)c.r * S.total_slots;
+  for (int d = 0; d < S.n_dc; ++d) {
+    int lo = S.slot_off[d], hi = S.slot_off[d + 1];
+    int cnt_inf = 0;
+    for (int k = lo + c.lane; k < hi; k += 64)
+      if (S.s_gpus[base + k] != 0 && S.s_jtype[base + k] == 0) cnt_inf++;
+    cnt_inf = wave_sum_i32(cnt_inf);
+    int rd = c.r * S.n_dc + d;
+    int run_total = S.n_running[rd];
+    if (c.lane == 0) {
+      int idx = *S.cl_count;
+      if (idx < S.cl_cap) {
+        double* row = &S.cl_rows[(int64_t)idx * 16];
+        double begin = S.util_begin[rd];
+        double elapsed = fmax(1e-9, now - (begin >= 0 ? begin : now));
+        row[0] = now;
+        row[1] = d;
+        row[2] = S.cur_freq[rd];
+        row[3] = S.busy[rd];
+        row[4] = S.total_gpus[d] - S.busy[rd];
+        row[5] = run_total;
+        row[6] = cnt_inf;
+        row[7] = run_total - cnt_inf;
+        row[8] = S.q_len[rd * 2 + 0];
+        row[9] = S.q_len[rd * 2 + 1];
+        row[10] = S.total_gpus[d] ? (double)S.busy[rd] / S.total_gpus[d] : 0.0;
+        row[11] = S.total_gpus[d]
+            ? S.util_time[rd] / (S.total_gpus[d] * elapsed) : 0.0;
+        row[12] = S.acc_unit[rd];
+        row[13] = c.dc_power(d);
+        row[14] = S.energy_j[rd];
+        row[15] = 0.0;
+        *S.cl_count = idx + 1;
+      } else {
+        atomicOr(&S.err[c.r], ERR_LOG_OVF);
+      }
+    }
+    store_fence();
+  }
+}
+
+__device__ void emit_job_row(Ctx& c, int d, int jt, int jid, int ing,
+                             float size, double fused, int n, float netlat,
+                             double start, double finish) {
+  const EngineDesc& S = *c.S;
+  if (c.r != S.log_replica) return;
+  if (c.lane == 0) {
+    int idx = *S.jl_count;
+    if (idx < S.jl_cap) {
+      double* row = &S.jl_rows[(int64_t)idx * 10];
+      row[0] = jid; row[1] = ing; row[2] = jt; row[3] = size; row[4] = d;
+      row[5] = fused; row[6] = n; row[7] = netlat; row[8] = start;
+      row[9] = finish;
+      *S.jl_count = idx + 1;
+    } else {
+      atomicOr(&S.err[c.r], ERR_LOG_OVF);
+    }
+  }
+  store_fence();
+}
+
+// cap_greedy controller step at log ticks (reference :248-315, v1: repeatedly
+// apply the currently-cheapest single down-step atom with exact re-estimation;
+// distributionally equivalent to the reference's per-pass sorted ladders)
+__device__ void cap_greedy_control(Ctx& c, double now) {
+  const EngineDesc& S = *c.S;
+  double f_min = S.freq_levels[0];
+  for (int k = 1; k < S.n_freq; ++k) f_min = fmin(f_min, S.freq_levels[k]);
+  int guard = 10000;
+  while (guard-- > 0) {
+    double totalP = 0;
+    for (int d = 0; d < S.n_dc; ++d) totalP += c.dc_power(d);
+    double deficit = totalP - S.power_cap;
+    if (deficit <= 1e-6) break;
+    // find min-rho single-step-down atom over all running jobs (lane-strided)
+    int64_t base = (int64_t)c.r * S.total_slots;
+    double best_rho = D_INF;
+    int best_slot = -1;
+    double best_fto = 0;
+    for (int k = c.lane; k < S.total_slots; k += 64) {
+      if (S.s_gpus[base + k] == 0) continue;
+      int d = S.slot_dc[k];
+      double fu = S.s_fused[base + k];
+      if (fu <= f_min + 1e-12) continue;
+      // nearest ladder index
+      int i0 = 0;
+      double bd = 1e300;
+      for (int q = 0; q < S.n_freq; ++q) {
+        double diff = fabs(S.freq_levels[q] - fu);
+        if (diff < bd) { bd = diff; i0 = q; }
+      }
+      if (i0 == 0) continue;
+      int jt = S.s_jtype[base + k];
+      int n = S.s_gpus[base + k];
+      double f_to = S.freq_levels[i0 - 1];
+      double v1 = 1.0 / d_unit_time(n, S.freq_levels[i0], c.lc3(d, jt));
+      double p1 = d_job_power(n, S.freq_levels[i0], c.pc3(d, jt));
+      double v2 = 1.0 / d_unit_time(n, f_to, c.lc3(d, jt));
+      double p2 = d_job_power(n, f_to, c.pc3(d, jt));
+      double dV = fmax(0.0, v1 - v2), dP = fmax(0.0, p1 - p2);
+      if (dV <= 0 || dP < 0) continue;
+      double rho = dP / dV;
+      if (rho < best_rho) { best_rho = rho; best_slot = k; best_fto = f_to; }
+    }
+    int wl;
+    double rho = wave_argmin_f64(best_rho, wl);
+    if (rho >= D_INF) break;
+    best_slot = __shfl(best_slot, wl, 64);
+    best_fto = __shfl(best_fto, wl, 64);
+    // apply: advance progress implicitly by recomputing remaining time
+    int d = S.slot_dc[best_slot];
+    int jt = S.s_jtype[base + best_slot];
+    int n = S.s_gpus[base + best_slot];
+    double old_f = S.s_fused[base + best_slot];
+    double T_old = d_unit_time(n, old_f, c.lc3(d, jt));
+    double T_new = d_unit_time(n, best_fto, c.lc3(d, jt));
+    double finish_old = S.s_finish[base + best_slot];
+    double remaining_units = fmax(0.0, (finish_old - now)) / fmax(T_old, 1e-12);
+    double finish_new = now + remaining_units * T_new;
+    if (c.lane == 0) {
+      int rd = c.r * S.n_dc + d;
+      S.p_active[rd] += d_job_power(n, best_fto, c.pc3(d, jt)) -
+                        d_job_power(n, old_f, c.pc3(d, jt));
+      S.sum_tpt[rd] += 1.0 / T_new - 1.0 / T_old;
+      S.s_fused[base + best_slot] = (float)best_fto;
+      S.s_finish[base + best_slot] = finish_new;
+    }
+    store_fence();
+    rescan_dc_min(c, d);
+  }
+}
+
+// ---------------- the advance kernel ----------------
+template <int ALGO>
+__global__ void __launch_bounds__(256)
+advance_kernel(EngineDesc S, double t_target, long long max_ev) {
+  int wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  int lane = threadIdx.x & 63;
+  if (wave >= S.n_rep) return;
+
+  Ctx c;
+  c.S = &S;
+  c.r = wave;
+  c.lane = lane;
+  if (S.done[c.r]) return;
+  c.now = S.now[c.r];
+  c.rng.key = S.seed ^ (0x9E3779B97F4A7C15ull * (uint64_t)(S.rep_id_offset + c.r));
+  c.rng.ctr = S.rng_ctr[c.r];
+
+  const int NS = S.n_streams;
+  int64_t sbase = (int64_t)c.r * S.total_slots;
+  long long n_events = 0;
+
+  while (n_events < max_ev) {
+    // ---- 1. next event: wave argmin over candidate sources ----
+    // per-lane candidate: value + kind/idx
+    double v = D_INF;
+    int kind = -1, idx = -1;
+    // arrival streams on lanes [0, NS)
+    if (lane < NS) {
+      double t = S.arr_next[(int64_t)c.r * NS + lane];
+      if (t < v) { v = t; kind = 0; idx = lane; }
+    }
+    // log tick on lane NS
+    if (lane == NS) {
+      double t = S.next_log[c.r];
+      if (t < v) { v = t; kind = 3; idx = 0; }
+    }
+    // dc min finishes on lanes [32, 32+n_dc)
+    if (lane >= 32 && lane < 32 + S.n_dc) {
+      double t = S.dc_min_finish[c.r * S.n_dc + (lane - 32)];
+      if (t < v) { v = t; kind = 2; idx = S.dc_min_slot[c.r * S.n_dc + (lane - 32)]; }
+    }
+    // transfers: strided over tcap
+    for (int k = lane; k < S.tcap; k += 64) {
+      double t = S.x_time[(int64_t)c.r * S.tcap + k];
+      if (t < v) { v = t; kind = 1; idx = k; }
+    }
+    int wl;
+    double t_min = wave_argmin_f64(v, wl);
+    kind = __shfl(kind, wl, 64);
+    idx = __shfl(idx, wl, 64);
+
+    if (t_min > S.end_time) {
+      // ---- end of simulation for this replica: final flush ----
+      if (lane < S.n_dc) {
+        int rd = c.r * S.n_dc + lane;
+        double last = c.now;
+        if (last >= 0.0 && last < S.end_time) {
+          S.util_time[rd] += S.busy[rd] * (S.end_time - last);
+          // reference quirk: the final accrue_energy(end) uses the BASELINE
+          // idle/sleep + p_peak*f^alpha model (models.py:82-91)
+          double f = S.cur_freq[rd];
+          int active = S.busy[rd];
+          int idlec = S.total_gpus[lane] - active;
+          double pa = active * (S.p_idle[lane] +
+                                S.p_peak[lane] * pow(f, S.pow_alpha[lane]));
+          double pi = idlec * (S.power_gating[lane] ? S.p_sleep[lane] : S.p_idle[lane]);
+          S.energy_j[rd] += (pa + pi) * (S.end_time - last);
+        }
+      }
+      if (lane == 0) S.done[c.r] = 1;
+      store_fence();
+      break;
+    }
+    if (t_min > t_target) break;  // chunk boundary
+
+    // ---- 2. accrue energy + util to t_min ----
+    accrue_to(c, t_min);
+    c.now = t_min;
+    n_events++;
+
+    // ---- 3. dispatch ----
+    if (kind == 0) {
+      // ===== arrival at ingress stream idx =====
+      int ing = idx >> 1;
+      int jt = idx & 1;
+      int jid = S.jid_ctr[c.r] + 1;
+      if (lane == 0) S.jid_ctr[c.r] = jid;
+      double size = jt == 0 ? rpareto_inf(c.rng)
+                            : fmax(0.1, rlognormal(c.rng, log(50000.0), 0.4));
+      // routing
+      int d_sel;
+      if (ALGO == A_ECO_ROUTE) {
+        // lanes 0..n_dc-1 each score their DC (serial 64-candidate loop)
+        double score = D_INF;
+        if (lane < S.n_dc) {
+          int d = lane;
+          const double* pcf = c.pc3(d, jt);
+          const double* tcf = c.lc3(d, jt);
+          double best = D_INF;
+          for (int n = 1; n <= S.max_gpj; ++n)
+            for (int q = 0; q < S.n_freq; ++q) {
+              double f = S.freq_levels[q];
+              double T = d_unit_time(n, f, tcf);
+              double E = d_job_power(n, f, pcf) * T;
+              double sc;
+              if (S.eco_obj == 1) sc = E * S.carbon[d];
+              else if (S.eco_obj == 2) sc = (E / 3.6e6) * c.price_kwh(t_min);
+              else sc = E;
+              if (sc < best) best = sc;
+            }
+          if (S.eco_obj == 1) score = (best * size);          // relative order ok
+          else if (S.eco_obj == 2) score = (best * size);
+          else score = best * size;
+        }
+        int dl;
+        wave_argmin_f64(score, dl);
+        d_sel = dl;
+      } else {
+        d_sel = (int)rbelow(c.rng, (uint32_t)S.n_dc);
+      }
+      double lnet = S.wan_lat[ing * S.n_dc + d_sel];
+      double bw = S.wan_bw[ing * S.n_dc + d_sel];
+      double xfer = bw > 0.0 ? S.payload_gb[jt] / bw : 0.0;
+      // push transfer record
+      int cand = INT_MAX;
+      for (int k = lane; k < S.tcap; k += 64) {
+        if (S.x_time[(int64_t)c.r * S.tcap + k] >= D_INF) { cand = k; break; }
+      }
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1)
+        cand = min(cand, __shfl_xor(cand, off, 64));
+      if (cand == INT_MAX) {
+        if (lane == 0) atomicOr(&S.err[c.r], ERR_XFER_OVF);
+      } else if (lane == 0) {
+        int64_t at = (int64_t)c.r * S.tcap + cand;
+        S.x_time[at] = t_min + lnet + xfer;
+        S.x_size[at] = (float)size;
+        S.x_netlat[at] = (float)lnet;
+        S.x_jid[at] = jid;
+        S.x_dc[at] = (char)d_sel;
+        S.x_jtype[at] = (char)jt;
+        S.x_ing[at] = (char)ing;
+      }
+      store_fence();
+      // next arrival for this stream (faithful non-accumulating thinning;
+      // reference arrivals.py:35-48)
+      double ia = D_INF;
+      {
+        double rate = S.arr_rate[jt];
+        int mode = S.arr_mode[jt];
+        double amp = S.arr_amp[jt];
+        double period = S.arr_period[jt];
+        if (mode == 0 && rate > 0) {
+          ia = rexp(c.rng, rate);
+        } else if (mode == 1) {
+          double max_rate = rate * (1.0 + fabs(amp));
+          if (max_rate > 0) {
+            for (int it = 0; it < 4096; ++it) {
+              double w = rexp(c.rng, max_rate);
+              double lam = fmax(0.0, rate * (1.0 + amp *
+                  sin(2.0 * M_PI * fmod(t_min + w, period) / period)));
+              if (u01(c.rng) <= lam / max_rate) { ia = w; break; }
+            }
+          }
+        }
+      }
+      if (lane == 0) S.arr_next[(int64_t)c.r * NS + idx] = t_min + ia;
+      store_fence();
+
+    } else if (kind == 1) {
+      // ===== WAN transfer complete: admission =====
+      int64_t at = (int64_t)c.r * S.tcap + idx;
+      int d = S.x_dc[at];
+      int jt = S.x_jtype[at];
+      float size = S.x_size[at];
+      float netlat = S.x_netlat[at];
+      int jid = S.x_jid[at];
+      int ing = S.x_ing[at];
+      if (lane == 0) S.x_time[at] = D_INF;
+      store_fence();
+      if (c.free_gpus(d) > 0) {
+        int n; double f;
+        decide_nf<ALGO>(c, d, jt, size, t_min, n, f);
+        n = max(1, min(n, c.free_gpus(d)));
+        start_job(c, d, jt, size, netlat, jid, ing, n, f, t_min);
+      } else {
+        queue_push(c, d, jt, size, netlat, jid, ing);
+      }
+
+    } else if (kind == 2) {
+      // ===== job finish =====
+      int slot = idx;
+      int d = S.slot_dc[slot];
+      int64_t at = sbase + slot;
+      int jt = S.s_jtype[at];
+      int n = S.s_gpus[at];
+      float size = S.s_size[at];
+      double fused = S.s_fused[at];
+      double start = S.s_start[at];
+      float netlat = S.s_netlat[at];
+      int jid = S.s_jid[at];
+      int ingr = S.s_ing[at];
+      double T = d_unit_time(n, fused, c.lc3(d, jt));
+      if (lane == 0) {
+        int rd = c.r * S.n_dc + d;
+        S.s_finish[at] = D_INF;
+        S.s_gpus[at] = 0;
+        S.busy[rd] = max(0, S.busy[rd] - n);
+        S.n_running[rd] -= 1;
+        S.p_active[rd] -= d_job_power(n, fused, c.pc3(d, jt));
+        S.sum_tpt[rd] -= 1.0 / T;
+        // remainder job-units: window = finish mod log_interval (quirk)
+        S.acc_unit[rd] += (1.0 / T) * fmod(t_min, S.log_interval);
+        // metrics
+        S.jobs_done[c.r] += 1;
+        S.sum_lat[c.r] += t_min - start;
+        S.sum_wait[c.r] += 0.0;
+        if (jt == 0) {
+          S.jobs_done_inf[c.r] += 1;
+          S.sum_lat_inf[c.r] += t_min - start;
+        }
+        if (ALGO == A_BANDIT) {
+          // reward = -E_pred (energy per unit at the used f)
+          double E = d_job_power(n, fused, c.pc3(d, jt)) * T;
+          int64_t ab = ((int64_t)c.r * S.n_dc + d) * 2 + jt;
+          for (int k = 0; k < S.n_freq; ++k)
+            if (S.freq_levels[k] == fused) {
+              S.b_n[ab * S.n_freq + k] += 1;
+              S.b_s[ab * S.n_freq + k] += (float)(-E);
+              break;
+            }
+        }
+      }
+      store_fence();
+      emit_job_row(c, d, jt, jid, ingr, size, fused, n, netlat, start, t_min);
+      rescan_dc_min(c, d);
+      drain_queues<ALGO>(c, d, t_min);
+      rescan_dc_min(c, d);
+
+    } else {
+      // ===== log tick =====
+      // power-cap control first (reference :463-465 order)
+      if (S.power_cap > 0) {
+        if (ALGO == A_CAP_GREEDY) {
+          cap_greedy_control(c, t_min);
+        } else if (ALGO == A_ECO_ROUTE || ALGO == A_CARBON_COST) {
+          if (lane < S.n_dc) {
+            int rd = c.r * S.n_dc + lane;
+            if (S.busy[rd] == 0) {
+              double fm = S.freq_levels[0];
+              for (int k = 1; k < S.n_freq; ++k) fm = fmin(fm, S.freq_levels[k]);
+              S.cur_freq[rd] = (float)fm;
+            }
+          }
+          store_fence();
+        }
+        // cap_uniform: exact no-op (per-job f_used power model; see oracle)
+      }
+      // acc_job_unit for running jobs: cached sum_tpt * interval
+      if (lane < S.n_dc) {
+        int rd = c.r * S.n_dc + lane;
+        S.acc_unit[rd] += S.sum_tpt[rd] * S.log_interval;
+      }
+      store_fence();
+      emit_cluster_rows(c, t_min);
+      if (lane == 0) S.next_log[c.r] = t_min + S.log_interval;
+      store_fence();
+    }
+  }
+
+  if (lane == 0) {
+    S.now[c.r] = c.now;
+    S.rng_ctr[c.r] = c.rng.ctr;
+    S.ev_count[c.r] += n_events;
+  }
+}
+
+// ---------------- host side ----------------
+#define T_PTR(name, type) S_.name = t_[#name].data_ptr<type>()
+#define T_CPTR(name, type) S_.name = t_[#name].data_ptr<type>()
+
+class BatchedSimHip {
+ public:
+  BatchedSimHip(py::dict tensors, py::dict cfg) {
+    // keep tensor refs alive
+    for (auto item : tensors)
+      t_[py::cast<std::string>(item.first)] = py::cast<torch::Tensor>(item.second);
+
+    S_.n_rep = cfg["n_rep"].cast<int>();
+    S_.n_dc = cfg["n_dc"].cast<int>();
+    S_.n_ing = cfg["n_ing"].cast<int>();
+    S_.n_freq = cfg["n_freq"].cast<int>();
+    S_.n_streams = S_.n_ing * 2;
+    S_.total_slots = cfg["total_slots"].cast<int>();
+    S_.tcap = cfg["tcap"].cast<int>();
+    S_.qcap = cfg["qcap"].cast<int>();
+    S_.end_time = cfg["end_time"].cast<double>();
+    S_.log_interval = cfg["log_interval"].cast<double>();
+    S_.algo = cfg["algo"].cast<int>();
+    S_.max_gpj = cfg["max_gpj"].cast<int>();
+    S_.inf_priority = cfg["inf_priority"].cast<int>();
+    S_.scale_out_low = cfg["scale_out_low"].cast<int>();
+    S_.energy_aware = cfg["energy_aware"].cast<int>();
+    S_.dvfs_low = cfg["dvfs_low"].cast<double>();
+    S_.dvfs_high = cfg["dvfs_high"].cast<double>();
+    S_.power_cap = cfg["power_cap"].cast<double>();
+    S_.eco_obj = cfg["eco_obj"].cast<int>();
+    S_.num_fixed = cfg["num_fixed"].cast<int>();
+    S_.fixed_freq = cfg["fixed_freq"].cast<double>();
+    S_.payload_gb[0] = cfg["payload_inf_gb"].cast<double>();
+    S_.payload_gb[1] = cfg["payload_trn_gb"].cast<double>();
+    auto am = cfg["arr_mode"].cast<std::vector<int>>();
+    auto ar = cfg["arr_rate"].cast<std::vector<double>>();
+    auto aa = cfg["arr_amp"].cast<std::vector<double>>();
+    auto ap = cfg["arr_period"].cast<std::vector<double>>();
+    for (int j = 0; j < 2; ++j) {
+      S_.arr_mode[j] = am[j]; S_.arr_rate[j] = ar[j];
+      S_.arr_amp[j] = aa[j]; S_.arr_period[j] = ap[j];
+    }
+    S_.seed = cfg["seed"].cast<uint64_t>();
+    S_.rep_id_offset = cfg["rep_id_offset"].cast<int64_t>();
+    S_.log_replica = cfg["log_replica"].cast<int>();
+    S_.cl_cap = cfg["cl_cap"].cast<int>();
+    S_.jl_cap = cfg["jl_cap"].cast<int>();
+
+    T_CPTR(freq_levels, double); T_CPTR(pc, double); T_CPTR(lc, double);
+    T_CPTR(wan_lat, double); T_CPTR(wan_bw, double); T_CPTR(carbon, double);
+    T_CPTR(price24, double); T_CPTR(total_gpus, int); T_CPTR(p_idle, double);
+    T_CPTR(p_sleep, double); T_CPTR(p_peak, double); T_CPTR(pow_alpha, double);
+    T_CPTR(power_gating, int); T_CPTR(default_freq, double);
+    T_CPTR(slot_off, int); T_CPTR(slot_dc, int);
+    T_PTR(now, double); T_PTR(next_log, double);
+    S_.rng_ctr = reinterpret_cast<uint64_t*>(t_["rng_ctr"].data_ptr<int64_t>());
+    T_PTR(jid_ctr, int); T_PTR(done, int); T_PTR(err, int);
+    T_PTR(arr_next, double);
+    T_PTR(busy, int); T_PTR(cur_freq, float); T_PTR(energy_j, double);
+    T_PTR(util_time, double); T_PTR(util_begin, double); T_PTR(acc_unit, double);
+    T_PTR(p_active, double); T_PTR(sum_tpt, double); T_PTR(n_running, int);
+    T_PTR(dc_min_finish, double); T_PTR(dc_min_slot, int);
+    T_PTR(s_finish, double); T_PTR(s_start, double); T_PTR(s_size, float);
+    T_PTR(s_fused, float); T_PTR(s_netlat, float); T_PTR(s_jid, int);
+    S_.s_gpus = reinterpret_cast<short*>(t_["s_gpus"].data_ptr<int16_t>());
+    S_.s_jtype = reinterpret_cast<char*>(t_["s_jtype"].data_ptr<int8_t>());
+    S_.s_ing = reinterpret_cast<char*>(t_["s_ing"].data_ptr<int8_t>());
+    T_PTR(x_time, double); T_PTR(x_size, float); T_PTR(x_netlat, float);
+    T_PTR(x_jid, int);
+    S_.x_dc = reinterpret_cast<char*>(t_["x_dc"].data_ptr<int8_t>());
+    S_.x_jtype = reinterpret_cast<char*>(t_["x_jtype"].data_ptr<int8_t>());
+    S_.x_ing = reinterpret_cast<char*>(t_["x_ing"].data_ptr<int8_t>());
+    T_PTR(q_head, int); T_PTR(q_len, int); T_PTR(q_size, float);
+    T_PTR(q_netlat, float); T_PTR(q_jid, int);
+    S_.q_ing = reinterpret_cast<char*>(t_["q_ing"].data_ptr<int8_t>());
+    T_PTR(b_n, int); T_PTR(b_s, float);
+    S_.b_t = reinterpret_cast<long long*>(t_["b_t"].data_ptr<int64_t>());
+    S_.ev_count = reinterpret_cast<long long*>(t_["ev_count"].data_ptr<int64_t>());
+    S_.jobs_done = reinterpret_cast<long long*>(t_["jobs_done"].data_ptr<int64_t>());
+    S_.jobs_done_inf =
+        reinterpret_cast<long long*>(t_["jobs_done_inf"].data_ptr<int64_t>());
+    T_PTR(sum_lat, double); T_PTR(sum_lat_inf, double); T_PTR(sum_wait, double);
+    T_PTR(cl_count, int); T_PTR(cl_rows, double);
+    T_PTR(jl_count, int); T_PTR(jl_rows, double);
+  }
+
+  // launch one advance chunk; returns immediately (stream-async)
+  void advance(double t_target, int64_t max_ev) {
+    int waves_per_block = 4;  // 256 threads
+    int blocks = (S_.n_rep + waves_per_block - 1) / waves_per_block;
+    dim3 grid(blocks), block(64 * waves_per_block);
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    switch (S_.algo) {
+      case A_DEFAULT:
+        hipLaunchKernelGGL(advance_kernel<A_DEFAULT>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      case A_CAP_UNIFORM:
+        hipLaunchKernelGGL(advance_kernel<A_CAP_UNIFORM>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      case A_CAP_GREEDY:
+        hipLaunchKernelGGL(advance_kernel<A_CAP_GREEDY>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      case A_JOINT_NF:
+        hipLaunchKernelGGL(advance_kernel<A_JOINT_NF>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      case A_BANDIT:
+        hipLaunchKernelGGL(advance_kernel<A_BANDIT>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      case A_CARBON_COST:
+        hipLaunchKernelGGL(advance_kernel<A_CARBON_COST>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      case A_ECO_ROUTE:
+        hipLaunchKernelGGL(advance_kernel<A_ECO_ROUTE>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      case A_DEBUG:
+        hipLaunchKernelGGL(advance_kernel<A_DEBUG>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      default:
+        throw std::runtime_error("unsupported algo for HIP engine");
+    }
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess)
+      throw std::runtime_error(std::string("advance_kernel launch failed: ") +
+                               hipGetErrorString(e));
+  }
+
+ private:
+  EngineDesc S_;
+  std::unordered_map<std::string, torch::Tensor> t_;
+};
+
+#undef T_PTR
+#undef T_CPTR
+
+}  // namespace dcg
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "Batched MI355X (gfx950) Monte-Carlo replica engine";
+  py::class_<dcg::BatchedSimHip>(m, "BatchedSimHip")
+      .def(py::init<py::dict, py::dict>())
+      .def("advance", &dcg::BatchedSimHip::advance,
+           py::arg("t_target"), py::arg("max_events"));
+}

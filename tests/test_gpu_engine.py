@@ -1,0 +1,240 @@
+"""MI355X batched-engine GPU tests (SURVEY §4 strategies (c)/(d)):
+distributional equivalence vs the scalar engines, deterministic-decision
+exactness (grid-search / debug pins), conservation checks, contract checks.
+All marked gpu — run via gpurun on a real MI355X."""
+import csv
+import io
+import json
+import math
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+torch = pytest.importorskip("torch")
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+pytestmark = pytest.mark.gpu
+
+needs_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                               reason="needs MI355X")
+
+
+def make_engine(algo="default_policy", replicas=128, duration=120.0,
+                out_dir=None, enable_logs=False, **kw):
+    from distributed_cluster_gpus_amd.configs.paper import (build_arrivals,
+                                                            paper_scenario)
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    sc = paper_scenario()
+    inf, trn = build_arrivals()
+    return BatchedEngine(sc, inf, trn, algo=algo, replicas=replicas,
+                         duration=duration, log_interval=5.0, out_dir=out_dir,
+                         seed=123, enable_logs=enable_logs, **kw)
+
+
+def native_population(algo, duration, seeds, **kw):
+    """Run the C++ scalar engine over several seeds; return per-seed stats."""
+    from distributed_cluster_gpus_amd.configs.paper import (build_arrivals,
+                                                            paper_scenario)
+    from distributed_cluster_gpus_amd.engine.native import NativeEngine
+    import tempfile
+    rows = []
+    for s in seeds:
+        sc = paper_scenario()
+        inf, trn = build_arrivals()
+        with tempfile.TemporaryDirectory() as td:
+            eng = NativeEngine(sc, inf, trn, algo=algo, duration=duration,
+                               log_interval=5.0, out_dir=td, seed=s, **kw)
+            st = eng.run()
+        rows.append(st)
+    return rows
+
+
+@needs_gpu
+def test_engine_runs_and_conserves():
+    eng = make_engine(replicas=256, duration=120.0)
+    st = eng.run()
+    assert st["events"] > 0
+    assert st["jobs_completed"] > 0
+    assert int(eng.t["err"].max().item()) == 0
+    # utilization integrals positive; busy never negative
+    assert int(eng.t["busy"].min().item()) >= 0
+    assert float(eng.t["energy_j"].min().item()) > 0  # sleep floor > 0
+    # every replica processed a similar order of events (same workload)
+    evs = eng.t["ev_count"].cpu().numpy()
+    assert evs.min() > 0.5 * evs.mean()
+
+
+@needs_gpu
+def test_distributional_match_vs_native_default_policy():
+    """GPU replica-population means must match the scalar C++ engine's
+    cross-seed distribution on total energy, completed jobs and mean latency
+    (batched-vs-scalar equivalence, distributional form)."""
+    duration = 120.0
+    nat = native_population("default_policy", duration, seeds=range(10))
+    nat_energy = np.array([r["total_energy_j"] for r in nat])
+    nat_jobs = np.array([r["jobs_completed"] for r in nat])
+
+    eng = make_engine(replicas=256, duration=duration)
+    st = eng.run()
+    gpu_energy = eng.t["energy_j"].sum(dim=1).cpu().numpy()
+    gpu_jobs = eng.t["jobs_done"].cpu().numpy().astype(float)
+
+    # population means within 3 combined standard errors
+    for g, n, name in ((gpu_energy, nat_energy, "energy"),
+                       (gpu_jobs, nat_jobs, "jobs")):
+        se = math.sqrt(n.std() ** 2 / len(n) + g.std() ** 2 / len(g))
+        assert abs(g.mean() - n.mean()) < 4 * se + 1e-9, \
+            f"{name}: gpu {g.mean():.4g} vs native {n.mean():.4g} (se {se:.3g})"
+
+
+@needs_gpu
+def test_distributional_match_joint_nf():
+    duration = 100.0
+    nat = native_population("joint_nf", duration, seeds=range(8))
+    nat_energy = np.array([r["total_energy_j"] for r in nat])
+    eng = make_engine(algo="joint_nf", replicas=192, duration=duration)
+    eng.run()
+    gpu_energy = eng.t["energy_j"].sum(dim=1).cpu().numpy()
+    se = math.sqrt(nat_energy.std() ** 2 / len(nat_energy) +
+                   gpu_energy.std() ** 2 / len(gpu_energy))
+    assert abs(gpu_energy.mean() - nat_energy.mean()) < 4 * se + 1e-9
+
+
+@needs_gpu
+def test_joint_nf_decisions_exact(tmp_path):
+    """joint_nf's (n, f) choice is deterministic per (dc, jtype): the GPU
+    grid-argmin must equal the CPU best_nf_grid exactly (when not clamped by
+    free-GPU pressure)."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.models.coeffs import LatencyCoeffs, PowerCoeffs
+    from distributed_cluster_gpus_amd.policies.gridsearch import best_nf_grid
+    out = str(tmp_path / "jn")
+    eng = make_engine(algo="joint_nf", replicas=8, duration=60.0,
+                      out_dir=out, enable_logs=True)
+    eng.run()
+    sc = paper_scenario()
+    expect = {}
+    for d, dc in enumerate(sc.dc_names):
+        for j, jt in enumerate(("inference", "training")):
+            pC = PowerCoeffs(*sc.power_coeffs[d, j, :])
+            tC = LatencyCoeffs(*sc.latency_coeffs[d, j, :])
+            n, f, *_ = best_nf_grid(8, list(sc.freq_levels), pC, tC,
+                                    objective="energy")
+            expect[(dc, jt)] = (n, f)
+    with open(os.path.join(out, "job_log.csv")) as fh:
+        rows = list(csv.DictReader(fh))
+    assert rows, "logging replica produced no job rows"
+    checked = 0
+    for r in rows:
+        n_exp, f_exp = expect[(r["dc"], r["type"])]
+        # n may be clamped down under GPU pressure; f must match exactly then
+        if int(r["n_gpus"]) == n_exp:
+            assert abs(float(r["f_used"]) - f_exp) < 1e-9
+            checked += 1
+    assert checked > len(rows) * 0.5
+
+
+@needs_gpu
+def test_debug_pins_nf(tmp_path):
+    out = str(tmp_path / "dbg")
+    eng = make_engine(algo="debug", replicas=4, duration=60.0, out_dir=out,
+                      enable_logs=True, num_fixed_gpus=2, fixed_freq=0.7)
+    eng.run()
+    with open(os.path.join(out, "job_log.csv")) as fh:
+        rows = list(csv.DictReader(fh))
+    assert rows
+    started_direct = [r for r in rows if r["n_gpus"] == "2"]
+    assert len(started_direct) > 0.9 * len(rows)
+    for r in started_direct:
+        assert r["f_used"] == "0.700"
+
+
+@needs_gpu
+def test_log_schema_matches_scalar(tmp_path):
+    """The batched engine's CSV outputs use the exact reference schemas."""
+    from distributed_cluster_gpus_amd.utils.csvlog import (CLUSTER_COLUMNS,
+                                                           JOB_COLUMNS)
+    out = str(tmp_path / "logs")
+    eng = make_engine(replicas=4, duration=40.0, out_dir=out, enable_logs=True)
+    eng.run()
+    with open(os.path.join(out, "cluster_log.csv")) as fh:
+        header = fh.readline().strip().split(",")
+    assert header == CLUSTER_COLUMNS
+    with open(os.path.join(out, "job_log.csv")) as fh:
+        header = fh.readline().strip().split(",")
+    assert header == JOB_COLUMNS
+    # util/energy sanity on logged rows
+    with open(os.path.join(out, "cluster_log.csv")) as fh:
+        for row in csv.DictReader(fh):
+            assert 0.0 <= float(row["util_inst"]) <= 1.0
+            assert 0.0 <= float(row["util_avg"]) <= 1.0001
+            assert float(row["power_W"]) > 0
+
+
+@needs_gpu
+def test_bandit_and_eco_route_run():
+    for algo in ("bandit", "eco_route", "carbon_cost", "cap_greedy"):
+        eng = make_engine(algo=algo, replicas=64, duration=60.0,
+                          power_cap=30000.0 if algo == "cap_greedy" else 0.0)
+        st = eng.run()
+        assert st["jobs_completed"] > 0, algo
+
+
+@needs_gpu
+def test_cap_greedy_reduces_power():
+    e1 = make_engine(algo="cap_greedy", replicas=64, duration=100.0,
+                     power_cap=0.0)
+    e1.run()
+    e2 = make_engine(algo="cap_greedy", replicas=64, duration=100.0,
+                     power_cap=30000.0)
+    e2.run()
+    assert float(e2.t["energy_j"].sum().item()) < float(e1.t["energy_j"].sum().item())
+
+
+@needs_gpu
+def test_replica_shard_rng_independent_of_world():
+    """Sharding must not change per-replica streams: replica k of a 2-'rank'
+    split equals replica k of the single-rank run (same global ids)."""
+    e_full = make_engine(replicas=8, duration=60.0)
+    e_full.run()
+    e_lo = make_engine(replicas=8, duration=60.0, rank=0, world=2)
+    e_lo.run()
+    e_hi = make_engine(replicas=8, duration=60.0, rank=1, world=2)
+    e_hi.run()
+    full_jobs = e_full.t["jobs_done"].cpu()
+    lo = e_lo.t["jobs_done"].cpu()
+    hi = e_hi.t["jobs_done"].cpu()
+    assert torch.equal(full_jobs[:4], lo)
+    assert torch.equal(full_jobs[4:], hi)
+    assert torch.allclose(e_full.t["energy_j"].sum(dim=1).cpu()[:4],
+                          e_lo.t["energy_j"].sum(dim=1).cpu())
+
+
+@needs_gpu
+def test_smoke_entry():
+    sys.path.insert(0, REPO)
+    import __graft_entry__ as ge
+    ge.smoke()
+
+
+@needs_gpu
+def test_bench_contract():
+    """bench.py default invocation must emit the driver's JSON contract."""
+    r = subprocess.run([sys.executable, os.path.join(REPO, "bench.py"),
+                        "--steps", "3", "--warmup", "1",
+                        "--replicas-per-gpu", "512", "--events-per-step", "200",
+                        "--with-rl", "0"],
+                       capture_output=True, text=True, timeout=900, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.strip().splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+              "dtype", "data", "config"):
+        assert k in out, k
+    assert out["value"] > 0 and out["n_gpus"] == 1
+    assert out["metric"] == "sim_events_per_sec"
