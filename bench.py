@@ -32,7 +32,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--replicas-per-gpu", type=int, default=4096)
-    p.add_argument("--events-per-step", type=int, default=1000)
+    p.add_argument("--events-per-step", type=int, default=4000)
     p.add_argument("--algo", type=str, default="default_policy")
     p.add_argument("--seed", type=int, default=123)
     p.add_argument("--with-rl", type=int, default=1,
