@@ -5,6 +5,8 @@ aggregate events/sec for several replica counts."""
 import json
 import sys
 import time
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
