@@ -54,7 +54,7 @@ class BatchedEngine:
                  logger=None, show_progress: bool = False,
                  device: Optional[torch.device] = None,
                  rank: int = 0, world: int = 1,
-                 tcap: int = 256, qcap: int = 2048,
+                 tcap: int = 256, qcap: int = 8192,
                  events_per_launch: int = 50000,
                  enable_logs: bool = True, **_unused_rl_kwargs):
         if algo not in ALGOS:
