@@ -98,7 +98,7 @@ def main():
     # duration generous enough that no replica reaches end_time mid-bench
     # (~150-200 events per simulated second per replica on this workload)
     duration = max(1200.0, total_steps * args.events_per_step / 100.0)
-    qcap = int(max(8192, 0.4 * duration))
+    qcap = int(max(24576, 8 * duration))
 
     sc = paper_scenario()
     inf, trn = build_arrivals()  # sinusoid 6/s amp .6 period 300; poisson 0.3/s

@@ -54,7 +54,7 @@ class BatchedEngine:
                  logger=None, show_progress: bool = False,
                  device: Optional[torch.device] = None,
                  rank: int = 0, world: int = 1,
-                 tcap: int = 256, qcap: int = 8192,
+                 tcap: int = 256, qcap: int = 24576,
                  events_per_launch: int = 50000,
                  enable_logs: bool = True, **_unused_rl_kwargs):
         if algo not in ALGOS:
@@ -160,9 +160,11 @@ class BatchedEngine:
         t["q_head"] = torch.zeros((R, n_dc, 2), **i32)
         t["q_len"] = torch.zeros((R, n_dc, 2), **i32)
         t["q_size"] = torch.zeros((R, n_dc, 2, qcap), **f32)
-        t["q_netlat"] = torch.zeros((R, n_dc, 2, qcap), **f32)
-        t["q_jid"] = torch.zeros((R, n_dc, 2, qcap), **i32)
-        t["q_ing"] = torch.zeros((R, n_dc, 2, qcap), **i8)
+        # queue aux fields (net latency / jid / ingress) are only consumed by
+        # the logging replica's job rows -> single-replica allocation
+        t["q_netlat"] = torch.zeros((n_dc, 2, qcap), **f32)
+        t["q_jid"] = torch.zeros((n_dc, 2, qcap), **i32)
+        t["q_ing"] = torch.zeros((n_dc, 2, qcap), **i8)
         nb = 1 if algo != "bandit" else R
         t["b_n"] = torch.zeros((nb, n_dc, 2, n_freq), **i32)
         t["b_s"] = torch.zeros((nb, n_dc, 2, n_freq), **f32)

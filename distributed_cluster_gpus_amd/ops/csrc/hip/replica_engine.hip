@@ -116,13 +116,14 @@ struct EngineDesc {
   char* x_dc;
   char* x_jtype;
   char* x_ing;
-  // queues [r][dc][2] ring of capacity qcap
+  // queues [r][dc][2] ring of capacity qcap; aux fields (netlat/jid/ing)
+  // exist only for the logging replica ([dc][2][qcap], no replica dim)
   int* q_head;
   int* q_len;
   float* q_size;                  // [r][dc][2][qcap]
-  float* q_netlat;
-  int* q_jid;
-  char* q_ing;
+  float* q_netlat;                // [dc][2][qcap] (log replica only)
+  int* q_jid;                     // [dc][2][qcap] (log replica only)
+  char* q_ing;                    // [dc][2][qcap] (log replica only)
   // bandit state [r][dc][2][n_freq]
   int* b_n;
   float* b_s;
@@ -331,9 +332,12 @@ __device__ bool queue_push(Ctx& c, int d, int jt, float size, float netlat,
   if (c.lane == 0) {
     int64_t at = (int64_t)q * S.qcap + pos;
     S.q_size[at] = size;
-    S.q_netlat[at] = netlat;
-    S.q_jid[at] = jid;
-    S.q_ing[at] = (char)ing;
+    if (c.r == S.log_replica) {
+      int64_t aux = ((int64_t)(d * 2 + jt)) * S.qcap + pos;
+      S.q_netlat[aux] = netlat;
+      S.q_jid[aux] = jid;
+      S.q_ing[aux] = (char)ing;
+    }
     S.q_len[q] = len + 1;
   }
   store_fence();
@@ -349,9 +353,14 @@ __device__ bool queue_pop(Ctx& c, int d, int jt, float& size, float& netlat,
   int pos = S.q_head[q];
   int64_t at = (int64_t)q * S.qcap + pos;
   size = S.q_size[at];
-  netlat = S.q_netlat[at];
-  jid = S.q_jid[at];
-  ing = S.q_ing[at];
+  if (c.r == S.log_replica) {
+    int64_t aux = ((int64_t)(d * 2 + jt)) * S.qcap + pos;
+    netlat = S.q_netlat[aux];
+    jid = S.q_jid[aux];
+    ing = S.q_ing[aux];
+  } else {
+    netlat = 0.0f; jid = 0; ing = 0;
+  }
   if (c.lane == 0) {
     S.q_head[q] = (pos + 1) % S.qcap;
     S.q_len[q] = len - 1;

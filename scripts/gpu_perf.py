@@ -14,7 +14,7 @@ from distributed_cluster_gpus_amd.configs.paper import build_arrivals, paper_sce
 from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
 
 
-def run_one(replicas, duration=1200.0, algo="default_policy", qcap=8192):
+def run_one(replicas, duration=1200.0, algo="default_policy", qcap=24576):
     sc = paper_scenario()
     inf, trn = build_arrivals()
     eng = BatchedEngine(sc, inf, trn, algo=algo, replicas=replicas,
