@@ -37,7 +37,7 @@ from .oracle import ALGOS
 
 _ALGO_IDS = {"default_policy": 0, "cap_uniform": 1, "cap_greedy": 2,
              "joint_nf": 3, "bandit": 4, "carbon_cost": 5, "eco_route": 6,
-             "debug": 7}
+             "debug": 7, "chsac_af": 8}
 _ECO_IDS = {"energy": 0, "carbon": 1, "cost": 2}
 INF = 1e300
 
@@ -56,13 +56,14 @@ class BatchedEngine:
                  rank: int = 0, world: int = 1,
                  tcap: int = 256, qcap: int = 24576,
                  events_per_launch: int = 50000,
-                 enable_logs: bool = True, **_unused_rl_kwargs):
+                 enable_logs: bool = True,
+                 sla_p99_ms: float = 500.0, energy_budget_j=None,
+                 rl_device: str = 'cuda', rl_batch: int = 256,
+                 rl_warmup: int = 1000, rl_buffer: int = 200000,
+                 rl_train_interval: int = 256, rl_agent=None,
+                 tr_cap: int = 262144, **_unused):
         if algo not in ALGOS:
             raise ValueError(f"unknown algo {algo!r}")
-        if algo == "chsac_af":
-            raise NotImplementedError(
-                "chsac_af on the batched engine lands with the RL-batch phase; "
-                "use --engine oracle for RL runs this round")
         if not torch.cuda.is_available():
             raise RuntimeError("BatchedEngine requires a ROCm GPU "
                                "(no silent CPU fallback)")
@@ -75,6 +76,7 @@ class BatchedEngine:
         self.events_per_launch = int(events_per_launch)
         self.out_dir = out_dir
         self.rl = None
+        self.replay = None
 
         from ..parallel.sharding import replica_shard
         shard = replica_shard(int(replicas), rank, world)
@@ -194,6 +196,84 @@ class BatchedEngine:
         t["arr_next"].copy_(torch.as_tensor(arr_np, dtype=torch.float64))
         t["rng_ctr"].fill_(n_ing * 2)  # host consumed one block per stream
 
+        # ===== CHSAC-AF (RL-in-the-loop) =====
+        self.is_rl = algo == "chsac_af"
+        obs_dim = 1 + 6 * n_dc
+        self.obs_dim = obs_dim
+        self.sla_p99_ms = float(sla_p99_ms)
+        self._rl_batch = int(rl_batch)
+        self._rl_warmup = int(rl_warmup)
+        self._rl_train_interval = int(rl_train_interval)
+        self.rl_updates = 0
+        if self.is_rl:
+            t["req_flag"] = torch.zeros(R, **i32)
+            t["req_obs"] = torch.zeros((R, obs_dim), **f32)
+            t["req_mdc"] = torch.zeros(R, **i32)
+            t["req_mg"] = torch.zeros(R, **i32)
+            t["resp_dc"] = torch.zeros(R, **i32)
+            t["resp_g"] = torch.zeros(R, **i32)
+            t["pend_kind"] = torch.zeros(R, **i32)
+            t["pend_size"] = torch.zeros(R, **f32)
+            t["pend_netlat"] = torch.zeros(R, **f32)
+            t["pend_jid"] = torch.zeros(R, **i32)
+            t["pend_ing"] = torch.zeros(R, **i32)
+            t["pend_jt"] = torch.zeros(R, **i32)
+            t["pend_dc"] = torch.zeros(R, **i32)
+            t["pend_from_inf"] = torch.zeros(R, **i32)
+            u8 = dict(dtype=torch.uint8, device=dev)
+            t["slot_s0"] = torch.zeros((R, total_slots, obs_dim), **f32)
+            t["slot_adc"] = torch.zeros((R, total_slots), **u8)
+            t["slot_ag"] = torch.zeros((R, total_slots), **u8)
+            t["slot_mdc"] = torch.zeros((R, total_slots), **u8)
+            t["slot_mg"] = torch.zeros((R, total_slots), **u8)
+            t["slot_has_rl"] = torch.zeros((R, total_slots), **u8)
+            t["slot_nrew"] = torch.zeros((R, total_slots), **u8)
+            t["x_s0"] = torch.zeros((R, tcap, obs_dim), **f32)
+            t["x_nsel"] = torch.zeros((R, tcap), **i16)
+            t["x_adc"] = torch.zeros((R, tcap), **u8)
+            t["x_ag"] = torch.zeros((R, tcap), **u8)
+            t["x_mdc"] = torch.zeros((R, tcap), **u8)
+            t["x_mg"] = torch.zeros((R, tcap), **u8)
+            t["x_has_rl"] = torch.zeros((R, tcap), **u8)
+            t["lat_hist"] = torch.zeros((R, 2, 64), **i32)
+            t["lat_count"] = torch.zeros((R, 2), **i64)
+            t["lat_sum"] = torch.zeros((R, 2), **f64)
+            t["tr_count"] = torch.zeros(1, **i32)
+            t["tr_s0"] = torch.zeros((tr_cap, obs_dim), **f32)
+            t["tr_s1"] = torch.zeros((tr_cap, obs_dim), **f32)
+            t["tr_adc"] = torch.zeros(tr_cap, **u8)
+            t["tr_ag"] = torch.zeros(tr_cap, **u8)
+            t["tr_r"] = torch.zeros(tr_cap, **f32)
+            t["tr_costs"] = torch.zeros((tr_cap, 3), **f32)
+            t["tr_mdc"] = torch.zeros(tr_cap, **u8)
+            t["tr_mg"] = torch.zeros(tr_cap, **u8)
+            # agent + replay on the same device
+            from ..rl.agent import CHSACAgentConfig, make_agent
+            from ..rl.replay import ReplayRing
+            constraints = {"latency_p99": self.sla_p99_ms}
+            if power_cap and power_cap > 0:
+                constraints["power"] = float(power_cap)
+            if energy_budget_j and energy_budget_j > 0:
+                constraints["energy_total"] = float(energy_budget_j)
+            constraints["gpu_over"] = 0.0
+            self._cost_names = ["latency_p99", "power", "gpu_over"]
+            if rl_agent is not None:
+                self.rl = rl_agent
+            else:
+                self.rl = make_agent(CHSACAgentConfig(
+                    obs_dim=obs_dim, n_dc=n_dc,
+                    n_g_choices=int(scenario.policy.max_gpus_per_job),
+                    constraints=constraints, device=str(dev)))
+            self.replay = ReplayRing(capacity=int(rl_buffer), obs_dim=obs_dim,
+                                     n_costs=3, cost_names=self._cost_names,
+                                     n_dc=n_dc,
+                                     n_g=int(scenario.policy.max_gpus_per_job),
+                                     device=str(dev), seed=seed)
+            # bit-expansion LUT for mask bytes -> bool vectors
+            ar = torch.arange(max(n_dc, int(scenario.policy.max_gpus_per_job)),
+                              device=dev)
+            self._bitpos = ar
+
         self.t = t
         self.arrival_inf, self.arrival_trn = arrival_inf, arrival_trn
 
@@ -220,6 +300,8 @@ class BatchedEngine:
             "seed": int(seed), "rep_id_offset": int(shard.start),
             "log_replica": self.log_replica,
             "cl_cap": cl_cap, "jl_cap": jl_cap,
+            "obs_dim": obs_dim, "sla_p99_ms": float(sla_p99_ms),
+            "tr_cap": int(tr_cap) if self.is_rl else 0,
         }
         self._sim = self._mod.BatchedSimHip(t, cfg)
         self.meter = ThroughputMeter()
@@ -284,15 +366,73 @@ class BatchedEngine:
             if err != 0:
                 raise RuntimeError(f"batched engine error flags: {err:#x} "
                                    f"(queue/transfer/slot/log overflow)")
+            if self.is_rl:
+                self._rl_service()
             if bool(t["done"].min().item() == 1):
                 break
-            if launches > 100000:
+            if launches > 1000000:
                 raise RuntimeError("batched engine failed to converge")
         self.meter.count = int(t["ev_count"].sum().item())
         self.meter.stop()
         if self.log_replica >= 0 and self.out_dir is not None:
             self._write_logs()
         return self.stats()
+
+    # ---------------- CHSAC host service ----------------
+    def _expand_mask(self, bytes_tensor, width):
+        """uint8/int bitmask tensor [B] -> bool [B, width]."""
+        b = bytes_tensor.to(torch.int64).unsqueeze(1)
+        return (b >> torch.arange(width, device=b.device).unsqueeze(0)) & 1 > 0
+
+    def _rl_service(self):
+        """Serve pending action requests with ONE batched policy forward,
+        drain completed transitions into the replay ring, and run SAC train
+        steps at the configured cadence (the reference trains once per job
+        completion, :803-807; batching across replicas changes that cadence
+        — documented; with 1 replica and rl_train_interval=1 the per-event
+        cadence is recovered)."""
+        t = self.t
+        pend = t["req_flag"] == 1
+        n_req = int(pend.sum().item())
+        if n_req > 0:
+            idx = pend.nonzero(as_tuple=True)[0]
+            obs = t["req_obs"][idx]
+            n_dc = self.sc.n_dc
+            n_g = int(self.sc.policy.max_gpus_per_job)
+            m_dc = self._expand_mask(t["req_mdc"][idx], n_dc)
+            m_g = self._expand_mask(t["req_mg"][idx], n_g)
+            # guard: a fully-false mask wedges the categorical; allow all
+            m_dc[m_dc.sum(dim=1) == 0] = True
+            m_g[m_g.sum(dim=1) == 0] = True
+            with torch.no_grad():
+                a = self.rl.select_action_batch(obs, m_dc, m_g)
+            t["resp_dc"][idx] = a["dc"].to(torch.int32)
+            t["resp_g"][idx] = a["g"].to(torch.int32)
+            t["req_flag"][idx] = 2  # REQ_READY
+        # transitions -> replay
+        n_tr = int(t["tr_count"].item())
+        if n_tr > 0:
+            n_tr = min(n_tr, int(t["tr_s0"].shape[0]))
+            costs = t["tr_costs"][:n_tr]
+            self.replay.add_batch(
+                s=t["tr_s0"][:n_tr], s_next=t["tr_s1"][:n_tr],
+                a_dc=t["tr_adc"][:n_tr].to(torch.long),
+                a_g=t["tr_ag"][:n_tr].to(torch.long),
+                r=t["tr_r"][:n_tr], costs=costs,
+                done=torch.ones(n_tr, device=self.device),
+                mask_dc=self._expand_mask(t["tr_mdc"][:n_tr], self.sc.n_dc),
+                mask_g=self._expand_mask(t["tr_mg"][:n_tr],
+                                         int(self.sc.policy.max_gpus_per_job)))
+            t["tr_count"].zero_()
+            self._tr_since_train = getattr(self, "_tr_since_train", 0) + n_tr
+            if self.replay.size >= self._rl_warmup:
+                steps = self._tr_since_train // self._rl_train_interval
+                steps = min(steps, 64)  # bound per-launch training work
+                for _ in range(steps):
+                    self.rl.train_step(self.replay.sample(self._rl_batch))
+                    self.rl_updates += 1
+                if steps:
+                    self._tr_since_train = 0
 
     def stats(self):
         t = self.t
@@ -302,7 +442,7 @@ class BatchedEngine:
             "events": int(t["ev_count"].sum().item()),
             "wall_s": self.meter.elapsed_s,
             "events_per_sec": self.meter.per_sec,
-            "rl_updates": 0,
+            "rl_updates": self.rl_updates,
             "jobs_completed": jobs,
             "jobs_completed_inf": jobs_inf,
             "replicas": self.R,

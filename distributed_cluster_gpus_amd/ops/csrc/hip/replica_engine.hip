@@ -41,7 +41,14 @@ constexpr int MAX_DC = 8;
 constexpr int MAX_FREQ = 8;
 
 enum Algo { A_DEFAULT = 0, A_CAP_UNIFORM, A_CAP_GREEDY, A_JOINT_NF, A_BANDIT,
-            A_CARBON_COST, A_ECO_ROUTE, A_DEBUG };
+            A_CARBON_COST, A_ECO_ROUTE, A_DEBUG, A_CHSAC };
+
+// CHSAC-AF action-request state machine kinds
+enum PendKind : int { PEND_NONE = 0, PEND_ARRIVAL = 1, PEND_DRAIN = 2 };
+// request flags
+enum ReqFlag : int { REQ_IDLE = 0, REQ_PENDING = 1, REQ_READY = 2 };
+
+constexpr int LAT_BINS = 64;  // log10 bins over sojourn [1e-4 s, 1e4 s)
 
 enum Err : int { ERR_NONE = 0, ERR_QUEUE_OVF = 1, ERR_XFER_OVF = 2,
                  ERR_SLOT_OVF = 4, ERR_LOG_OVF = 8 };
@@ -142,6 +149,57 @@ struct EngineDesc {
   double* cl_rows;                // [cl_cap][16]
   int* jl_count;                  // [1]
   double* jl_rows;                // [jl_cap][10]
+  // ===== CHSAC-AF (RL-in-the-loop) state; null unless algo == A_CHSAC =====
+  int obs_dim;                    // 1 + 6*n_dc
+  double sla_p99_ms;
+  // action request/response, one slot per replica
+  int* req_flag;                  // [r] ReqFlag
+  float* req_obs;                 // [r][obs_dim]
+  int* req_mdc;                   // [r] bitmask of valid DCs
+  int* req_mg;                    // [r] bitmask of valid g choices
+  int* resp_dc;                   // [r]
+  int* resp_g;                    // [r]
+  int* pend_kind;                 // [r] PendKind
+  // stashed context for the paused event
+  float* pend_size;               // [r]
+  float* pend_netlat;             // [r]
+  int* pend_jid;                  // [r]
+  int* pend_ing;                  // [r]
+  int* pend_jt;                   // [r]
+  int* pend_dc;                   // [r] (drain: source DC)
+  int* pend_from_inf;             // [r] (drain: which queue the job came from)
+  // per-job RL traces (state0 / action / masks at selection time)
+  float* slot_s0;                 // [r][total_slots][obs_dim]
+  unsigned char* slot_adc;        // [r][total_slots]
+  unsigned char* slot_ag;         // [r][total_slots]
+  unsigned char* slot_mdc;        // [r][total_slots]
+  unsigned char* slot_mg;         // [r][total_slots]
+  unsigned char* slot_has_rl;     // [r][total_slots]
+  unsigned char* slot_nrew;       // [r][total_slots] n used in the reward
+                                  // (arrival path: g_idx+1 unclamped, drain
+                                  // path: clamped — reference :571 vs :889)
+  float* x_s0;                    // [r][tcap][obs_dim]
+  short* x_nsel;                  // [r][tcap] RL-chosen n (before clamping)
+  unsigned char* x_adc;           // [r][tcap]
+  unsigned char* x_ag;            // [r][tcap]
+  unsigned char* x_mdc;           // [r][tcap]
+  unsigned char* x_mg;            // [r][tcap]
+  unsigned char* x_has_rl;        // [r][tcap]
+  // latency histograms per (r, jtype): counts in log10 bins
+  int* lat_hist;                  // [r][2][LAT_BINS]
+  long long* lat_count;           // [r][2]
+  double* lat_sum;                // [r][2]
+  // transition ring (global across replicas; host drains between launches)
+  int tr_cap;
+  int* tr_count;                  // [1] atomicAdd cursor
+  float* tr_s0;                   // [cap][obs_dim]
+  float* tr_s1;                   // [cap][obs_dim]
+  unsigned char* tr_adc;          // [cap]
+  unsigned char* tr_ag;           // [cap]
+  float* tr_r;                    // [cap]
+  float* tr_costs;                // [cap][3]: latency_p99_ms, power_W, gpu_over
+  unsigned char* tr_mdc;          // [cap]
+  unsigned char* tr_mg;           // [cap]
 };
 
 // ---------------- wave helpers ----------------
@@ -344,6 +402,32 @@ __device__ bool queue_push(Ctx& c, int d, int jt, float size, float netlat,
   return true;
 }
 
+__device__ bool queue_push_front(Ctx& c, int d, int jt, float size,
+                                 float netlat, int jid, int ing) {
+  const EngineDesc& S = *c.S;
+  int q = (c.r * S.n_dc + d) * 2 + jt;
+  int len = S.q_len[q];
+  if (len >= S.qcap) {
+    if (c.lane == 0) atomicOr(&S.err[c.r], ERR_QUEUE_OVF);
+    return false;
+  }
+  int pos = (S.q_head[q] - 1 + S.qcap) % S.qcap;
+  if (c.lane == 0) {
+    int64_t at = (int64_t)q * S.qcap + pos;
+    S.q_size[at] = size;
+    if (c.r == S.log_replica) {
+      int64_t aux = ((int64_t)(d * 2 + jt)) * S.qcap + pos;
+      S.q_netlat[aux] = netlat;
+      S.q_jid[aux] = jid;
+      S.q_ing[aux] = (char)ing;
+    }
+    S.q_head[q] = pos;
+    S.q_len[q] = len + 1;
+  }
+  store_fence();
+  return true;
+}
+
 __device__ bool queue_pop(Ctx& c, int d, int jt, float& size, float& netlat,
                           int& jid, int& ing) {
   const EngineDesc& S = *c.S;
@@ -540,6 +624,213 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
   }
 }
 
+// ---------------- CHSAC-AF (RL) device helpers ----------------
+// obs vector [now] + per-DC [total, busy, free, current_f, q_inf, q_train]
+// (reference _upgr_obs, simulator_paper_multi.py:1041-1053); lanes 0..n_dc-1
+// write their DC's 6 features in parallel.
+__device__ void rl_build_obs(Ctx& c, double now, float* out) {
+  const EngineDesc& S = *c.S;
+  if (c.lane == 0) out[0] = (float)now;
+  if (c.lane < S.n_dc) {
+    int rd = c.r * S.n_dc + c.lane;
+    float total = (float)S.total_gpus[c.lane];
+    float busy = (float)S.busy[rd];
+    out[1 + 6 * c.lane + 0] = total;
+    out[1 + 6 * c.lane + 1] = busy;
+    out[1 + 6 * c.lane + 2] = fmaxf(0.0f, total - busy);
+    out[1 + 6 * c.lane + 3] = S.cur_freq[rd];
+    out[1 + 6 * c.lane + 4] = (float)S.q_len[rd * 2 + 0];
+    out[1 + 6 * c.lane + 5] = (float)S.q_len[rd * 2 + 1];
+  }
+  store_fence();
+}
+
+// approximate p99 (ms) from the per-replica log-binned latency histogram.
+// The reference computes an exact percentile over a 2048-sample sliding
+// window (:728-737); the histogram form is the batched approximation
+// (documented divergence; exact at the oracle).
+__device__ double rl_p99_ms(Ctx& c, int jt) {
+  const EngineDesc& S = *c.S;
+  long long total = S.lat_count[c.r * 2 + jt];
+  if (total < 5) return -1.0;
+  long long target = (long long)(0.99 * (double)total);
+  long long cum = 0;
+  const int* h = &S.lat_hist[(c.r * 2 + jt) * LAT_BINS];
+  for (int b = 0; b < LAT_BINS; ++b) {
+    cum += h[b];
+    if (cum > target) {
+      // upper edge of bin b: 10^(-4 + 8*(b+1)/LAT_BINS) seconds
+      double s = pow(10.0, -4.0 + 8.0 * (b + 1) / LAT_BINS);
+      return s * 1000.0;
+    }
+  }
+  return 1e7;
+}
+
+__device__ void rl_record_latency(Ctx& c, int jt, double sojourn_s) {
+  const EngineDesc& S = *c.S;
+  if (c.lane != 0) return;
+  double l = log10(fmax(sojourn_s, 1e-9));
+  int b = (int)((l + 4.0) / 8.0 * LAT_BINS);
+  b = max(0, min(LAT_BINS - 1, b));
+  S.lat_hist[(c.r * 2 + jt) * LAT_BINS + b] += 1;
+  S.lat_count[c.r * 2 + jt] += 1;
+  S.lat_sum[c.r * 2 + jt] += sojourn_s;
+}
+
+// masks (reference _upgr_masks :1055-1082): DC valid if it has free GPUs;
+// g in 1..N valid if <= max free anywhere; g capped to 1 when recent p99 is
+// comfortably under the SLA (training window preferred).
+__device__ void rl_build_masks(Ctx& c, int& mdc, int& mg) {
+  const EngineDesc& S = *c.S;
+  int m1 = 0, max_free = 0;
+  for (int d = 0; d < S.n_dc; ++d) {
+    int free = c.free_gpus(d);
+    if (free > 0) m1 |= (1 << d);
+    max_free = max(max_free, free);
+  }
+  int m2 = 0;
+  for (int g = 1; g <= S.max_gpj; ++g)
+    if (g <= max_free) m2 |= (1 << (g - 1));
+  int jt_buf = S.lat_count[c.r * 2 + 1] > 0 ? 1 : 0;
+  double p99 = rl_p99_ms(c, jt_buf);
+  if (p99 >= 0.0 && p99 < 0.9 * S.sla_p99_ms) {
+    m2 = 0;
+    if (1 <= max_free) m2 = 1;  // cap at a single GPU
+  }
+  mdc = m1;
+  mg = m2;
+}
+
+// write an action request and stash the paused-event context
+__device__ void rl_request(Ctx& c, int kind, double now, int jt, int ing,
+                           float size, float netlat, int jid, int src_dc,
+                           int from_inf) {
+  const EngineDesc& S = *c.S;
+  rl_build_obs(c, now, &S.req_obs[(int64_t)c.r * S.obs_dim]);
+  int mdc, mg;
+  rl_build_masks(c, mdc, mg);
+  if (c.lane == 0) {
+    S.req_mdc[c.r] = mdc;
+    S.req_mg[c.r] = mg;
+    S.pend_kind[c.r] = kind;
+    S.pend_size[c.r] = size;
+    S.pend_netlat[c.r] = netlat;
+    S.pend_jid[c.r] = jid;
+    S.pend_ing[c.r] = ing;
+    S.pend_jt[c.r] = jt;
+    S.pend_dc[c.r] = src_dc;
+    S.pend_from_inf[c.r] = from_inf;
+    S.req_flag[c.r] = REQ_PENDING;
+  }
+  store_fence();
+}
+
+// min n meeting the SLA at fixed f (reference _min_n_for_sla :1091-1096)
+__device__ int rl_min_n_for_sla(Ctx& c, int d, int jt, double size, double f) {
+  const EngineDesc& S = *c.S;
+  for (int n = 1; n <= S.max_gpj; ++n)
+    if (size * d_unit_time(n, f, c.lc3(d, jt)) * 1000.0 <= S.sla_p99_ms)
+      return n;
+  return S.max_gpj;
+}
+
+// energy-optimal f at n over the sorted ladder (deadline guard is vacuous:
+// reference jobs never carry deadlines — Job.deadline is always None)
+__device__ double rl_energy_freq(Ctx& c, int d, int jt, int n) {
+  const EngineDesc& S = *c.S;
+  return wave_energy_freq(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                          S.n_freq, n);
+}
+
+// start a job carrying an RL trace; returns chosen slot via start_job's path.
+// (duplicates start_job, then fills the rl trace of the slot just used)
+__device__ void rl_start_job(Ctx& c, int d, int jt, float size, float netlat,
+                             int jid, int ing, int n, double f, double now,
+                             const float* s0, int a_dc, int a_g,
+                             int mdc, int mg, int n_rew) {
+  const EngineDesc& S = *c.S;
+  int64_t base = (int64_t)c.r * S.total_slots;
+  int lo = S.slot_off[d], hi = S.slot_off[d + 1];
+  int cand = INT_MAX;
+  for (int k = lo + c.lane; k < hi; k += 64) {
+    if (S.s_gpus[base + k] == 0) { cand = k; break; }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    cand = min(cand, __shfl_xor(cand, off, 64));
+  if (cand == INT_MAX) {
+    if (c.lane == 0) atomicOr(&S.err[c.r], ERR_SLOT_OVF);
+    return;
+  }
+  double T = d_unit_time(n, f, c.lc3(d, jt));
+  double finish = now + (double)size * T;
+  // copy s0 trace (lane-parallel over obs_dim)
+  for (int k = c.lane; k < S.obs_dim; k += 64)
+    S.slot_s0[(base + cand) * S.obs_dim + k] = s0[k];
+  if (c.lane == 0) {
+    S.s_finish[base + cand] = finish;
+    S.s_start[base + cand] = now;
+    S.s_size[base + cand] = size;
+    S.s_fused[base + cand] = (float)f;
+    S.s_netlat[base + cand] = netlat;
+    S.s_jid[base + cand] = jid;
+    S.s_gpus[base + cand] = (short)n;
+    S.s_jtype[base + cand] = (char)jt;
+    S.s_ing[base + cand] = (char)ing;
+    S.slot_adc[base + cand] = (unsigned char)a_dc;
+    S.slot_ag[base + cand] = (unsigned char)a_g;
+    S.slot_mdc[base + cand] = (unsigned char)mdc;
+    S.slot_mg[base + cand] = (unsigned char)mg;
+    S.slot_has_rl[base + cand] = 1;
+    S.slot_nrew[base + cand] = (unsigned char)max(1, n_rew);
+    int rd = c.r * S.n_dc + d;
+    S.busy[rd] += n;
+    S.n_running[rd] += 1;
+    S.p_active[rd] += d_job_power(n, f, c.pc3(d, jt));
+    S.sum_tpt[rd] += 1.0 / T;
+    if (finish < S.dc_min_finish[rd]) {
+      S.dc_min_finish[rd] = finish;
+      S.dc_min_slot[rd] = cand;
+    }
+  }
+  store_fence();
+}
+
+// emit a CHSAC transition (s0, s1, a, r, costs, masks) into the global ring
+__device__ void rl_emit_transition(Ctx& c, const float* s0, int a_dc, int a_g,
+                                   float r, float c_lat, float c_pow,
+                                   float c_over, int mdc, int mg, double now) {
+  const EngineDesc& S = *c.S;
+  // s1 = obs at completion time; build into a scratch row first (reuse the
+  // replica's request-obs row as scratch — safe: no pending request coexists
+  // with a finish emission in the same event)
+  float* s1 = &S.req_obs[(int64_t)c.r * S.obs_dim];
+  rl_build_obs(c, now, s1);
+  int idx = 0;
+  if (c.lane == 0) idx = atomicAdd(S.tr_count, 1);
+  idx = __shfl(idx, 0, 64);
+  if (idx >= S.tr_cap) {
+    if (c.lane == 0) atomicOr(&S.err[c.r], ERR_LOG_OVF);
+    return;
+  }
+  for (int k = c.lane; k < S.obs_dim; k += 64) {
+    S.tr_s0[(int64_t)idx * S.obs_dim + k] = s0[k];
+    S.tr_s1[(int64_t)idx * S.obs_dim + k] = s1[k];
+  }
+  if (c.lane == 0) {
+    S.tr_adc[idx] = (unsigned char)a_dc;
+    S.tr_ag[idx] = (unsigned char)a_g;
+    S.tr_r[idx] = r;
+    S.tr_costs[idx * 3 + 0] = c_lat;
+    S.tr_costs[idx * 3 + 1] = c_pow;
+    S.tr_costs[idx * 3 + 2] = c_over;
+    S.tr_mdc[idx] = (unsigned char)mdc;
+    S.tr_mg[idx] = (unsigned char)mg;
+  }
+  store_fence();
+}
+
 // ---------------- the advance kernel ----------------
 template <int ALGO>
 __global__ void __launch_bounds__(256)
@@ -560,6 +851,81 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   const int NS = S.n_streams;
   int64_t sbase = (int64_t)c.r * S.total_slots;
   long long n_events = 0;
+  bool paused = false;
+
+  // ---- CHSAC: resume a paused action request ----
+  if (ALGO == A_CHSAC && S.pend_kind[c.r] != PEND_NONE) {
+    if (S.req_flag[c.r] != REQ_READY) {
+      if (lane == 0) S.rng_ctr[c.r] = c.rng.ctr;
+      return;  // still waiting for the host policy
+    }
+    int a_dc = S.resp_dc[c.r];
+    int a_g = S.resp_g[c.r];
+    const float* s0 = &S.req_obs[(int64_t)c.r * S.obs_dim];
+    int mdc = S.req_mdc[c.r], mg = S.req_mg[c.r];
+    int pk = S.pend_kind[c.r];
+    int jt = S.pend_jt[c.r];
+    int ing = S.pend_ing[c.r];
+    float size = S.pend_size[c.r];
+    float netlat = S.pend_netlat[c.r];
+    int jid = S.pend_jid[c.r];
+    if (pk == PEND_ARRIVAL) {
+      // complete the arrival: RL chose (dc, g); push the WAN transfer
+      int d_sel = a_dc;
+      int n_sel = a_g + 1;
+      double lnet = S.wan_lat[ing * S.n_dc + d_sel];
+      double bw = S.wan_bw[ing * S.n_dc + d_sel];
+      double xfer = bw > 0.0 ? S.payload_gb[jt] / bw : 0.0;
+      int cand = INT_MAX;
+      for (int k = lane; k < S.tcap; k += 64) {
+        if (S.x_time[(int64_t)c.r * S.tcap + k] >= D_INF) { cand = k; break; }
+      }
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1)
+        cand = min(cand, __shfl_xor(cand, off, 64));
+      if (cand == INT_MAX) {
+        if (lane == 0) atomicOr(&S.err[c.r], ERR_XFER_OVF);
+      } else {
+        int64_t at = (int64_t)c.r * S.tcap + cand;
+        for (int k = lane; k < S.obs_dim; k += 64)
+          S.x_s0[at * S.obs_dim + k] = s0[k];
+        if (lane == 0) {
+          S.x_time[at] = c.now + lnet + xfer;
+          S.x_size[at] = size;
+          S.x_netlat[at] = (float)lnet;
+          S.x_jid[at] = jid;
+          S.x_dc[at] = (char)d_sel;
+          S.x_jtype[at] = (char)jt;
+          S.x_ing[at] = (char)ing;
+          S.x_nsel[at] = (short)n_sel;
+          S.x_adc[at] = (unsigned char)a_dc;
+          S.x_ag[at] = (unsigned char)a_g;
+          S.x_mdc[at] = (unsigned char)mdc;
+          S.x_mg[at] = (unsigned char)mg;
+          S.x_has_rl[at] = 1;
+        }
+      }
+    } else {  // PEND_DRAIN: one queued job, RL chose a target DC + g
+      int src_d = S.pend_dc[c.r];
+      int from_inf = S.pend_from_inf[c.r];
+      int d_tgt = a_dc;
+      if (c.free_gpus(d_tgt) <= 0) {
+        // no free target: push the job back at the FRONT of its source queue
+        // (reference :858-861)
+        queue_push_front(c, src_d, from_inf ? 0 : 1, size, netlat, jid, ing);
+      } else {
+        int n_sel = max(1, min(min(a_g + 1, c.free_gpus(d_tgt)), S.max_gpj));
+        double f = rl_energy_freq(c, d_tgt, jt, n_sel);
+        rl_start_job(c, d_tgt, jt, size, netlat, jid, ing, n_sel, f, c.now,
+                     s0, a_dc, a_g, mdc, mg, n_sel);
+      }
+    }
+    if (lane == 0) {
+      S.pend_kind[c.r] = PEND_NONE;
+      S.req_flag[c.r] = REQ_IDLE;
+    }
+    store_fence();
+  }
 
   while (n_events < max_ev) {
     // ---- 1. next event: wave argmin over candidate sources ----
@@ -629,6 +995,36 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       if (lane == 0) S.jid_ctr[c.r] = jid;
       double size = jt == 0 ? rpareto_inf(c.rng)
                             : fmax(0.1, rlognormal(c.rng, log(50000.0), 0.4));
+      if (ALGO == A_CHSAC) {
+        // schedule the next arrival first (RNG order differs from the scalar
+        // engines; distributionally identical), then pause for the policy
+        double ia_rl = D_INF;
+        {
+          double rate = S.arr_rate[jt];
+          int mode = S.arr_mode[jt];
+          double amp = S.arr_amp[jt];
+          double period = S.arr_period[jt];
+          if (mode == 0 && rate > 0) {
+            ia_rl = rexp(c.rng, rate);
+          } else if (mode == 1) {
+            double max_rate = rate * (1.0 + fabs(amp));
+            if (max_rate > 0) {
+              for (int it = 0; it < 4096; ++it) {
+                double w = rexp(c.rng, max_rate);
+                double lam = fmax(0.0, rate * (1.0 + amp *
+                    sin(2.0 * M_PI * fmod(t_min + w, period) / period)));
+                if (u01(c.rng) <= lam / max_rate) { ia_rl = w; break; }
+              }
+            }
+          }
+        }
+        if (lane == 0) S.arr_next[(int64_t)c.r * NS + idx] = t_min + ia_rl;
+        store_fence();
+        rl_request(c, PEND_ARRIVAL, t_min, jt, ing, (float)size, 0.0f, jid,
+                   -1, 0);
+        paused = true;
+        break;
+      }
       // routing
       int d_sel;
       if (ALGO == A_ECO_ROUTE) {
@@ -721,11 +1117,22 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       if (lane == 0) S.x_time[at] = D_INF;
       store_fence();
       if (c.free_gpus(d) > 0) {
-        int n; double f;
-        decide_nf<ALGO>(c, d, jt, size, t_min, n, f);
-        n = max(1, min(n, c.free_gpus(d)));
-        start_job(c, d, jt, size, netlat, jid, ing, n, f, t_min);
+        if (ALGO == A_CHSAC && S.x_has_rl[at]) {
+          // RL-chosen n (clamped), energy-optimal f (reference :646-667)
+          int n = max(1, min(min((int)S.x_nsel[at], c.free_gpus(d)), S.max_gpj));
+          double f = rl_energy_freq(c, d, jt, n);
+          rl_start_job(c, d, jt, size, netlat, jid, ing, n, f, t_min,
+                       &S.x_s0[at * S.obs_dim], S.x_adc[at], S.x_ag[at],
+                       S.x_mdc[at], S.x_mg[at], (int)S.x_nsel[at]);
+        } else {
+          int n; double f;
+          decide_nf<ALGO>(c, d, jt, size, t_min, n, f);
+          n = max(1, min(n, c.free_gpus(d)));
+          start_job(c, d, jt, size, netlat, jid, ing, n, f, t_min);
+        }
       } else {
+        // queueing drops the RL trace: a later chsac drain assigns a fresh
+        // observation/action (reference :849-889 overwrites them)
         queue_push(c, d, jt, size, netlat, jid, ing);
       }
 
@@ -776,8 +1183,54 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       store_fence();
       emit_job_row(c, d, jt, jid, ingr, size, fused, n, netlat, start, t_min);
       rescan_dc_min(c, d);
-      drain_queues<ALGO>(c, d, t_min);
-      rescan_dc_min(c, d);
+      if (ALGO == A_CHSAC) {
+        // record latency, then build the transition (reference :718-800)
+        double sojourn = fmax(0.0, t_min - start);
+        rl_record_latency(c, jt, sojourn);
+        store_fence();
+        if (S.slot_has_rl[at]) {
+          double E_pred = d_job_power(n, fused, c.pc3(d, jt)) * T;  // J/unit
+          double E_unit_kwh = (E_pred * (double)size / 3.6e6) /
+                              ((double)size + 1e-9);
+          int n_rew = max(1, (int)S.slot_nrew[at]);
+          float r = (float)(-E_unit_kwh + 0.05 * (1.0 / n_rew));
+          double p99 = rl_p99_ms(c, jt);
+          if (p99 < 0.0) p99 = sojourn * 1000.0;  // <5 samples fallback
+          double P_now = c.dc_power(d);
+          int n_min = rl_min_n_for_sla(c, d, jt, size, fused);
+          float c_over = (float)max(0, n - n_min);
+          int mdc, mg;
+          rl_build_masks(c, mdc, mg);  // masks at completion (reference :793)
+          rl_emit_transition(c, &S.slot_s0[at * S.obs_dim],
+                             S.slot_adc[at], S.slot_ag[at], r,
+                             (float)p99, (float)P_now, c_over, mdc, mg, t_min);
+          if (lane == 0) S.slot_has_rl[at] = 0;
+          store_fence();
+        }
+        // chsac drains AT MOST ONE queued job per finish, via a fresh policy
+        // action (reference :849-890)
+        float qsize, qnetlat;
+        int qjid, qing;
+        bool popped = false;
+        int from_inf = 0;
+        if (c.free_gpus(d) > 0) {
+          if (S.inf_priority && queue_pop(c, d, 0, qsize, qnetlat, qjid, qing)) {
+            popped = true;
+            from_inf = 1;
+          } else if (queue_pop(c, d, 1, qsize, qnetlat, qjid, qing)) {
+            popped = true;
+          }
+        }
+        if (popped) {
+          rl_request(c, PEND_DRAIN, t_min, from_inf ? 0 : 1, qing, qsize,
+                     qnetlat, qjid, d, from_inf);
+          paused = true;
+          break;
+        }
+      } else {
+        drain_queues<ALGO>(c, d, t_min);
+        rescan_dc_min(c, d);
+      }
 
     } else {
       // ===== log tick =====
@@ -901,6 +1354,42 @@ class BatchedSimHip {
     T_PTR(sum_lat, double); T_PTR(sum_lat_inf, double); T_PTR(sum_wait, double);
     T_PTR(cl_count, int); T_PTR(cl_rows, double);
     T_PTR(jl_count, int); T_PTR(jl_rows, double);
+
+    // CHSAC-AF extras
+    S_.obs_dim = cfg.contains("obs_dim") ? cfg["obs_dim"].cast<int>() : 0;
+    S_.sla_p99_ms = cfg.contains("sla_p99_ms") ? cfg["sla_p99_ms"].cast<double>() : 500.0;
+    S_.tr_cap = cfg.contains("tr_cap") ? cfg["tr_cap"].cast<int>() : 0;
+    if (S_.algo == A_CHSAC) {
+      T_PTR(req_flag, int); T_PTR(req_obs, float); T_PTR(req_mdc, int);
+      T_PTR(req_mg, int); T_PTR(resp_dc, int); T_PTR(resp_g, int);
+      T_PTR(pend_kind, int); T_PTR(pend_size, float); T_PTR(pend_netlat, float);
+      T_PTR(pend_jid, int); T_PTR(pend_ing, int); T_PTR(pend_jt, int);
+      T_PTR(pend_dc, int); T_PTR(pend_from_inf, int);
+      T_PTR(slot_s0, float);
+      S_.slot_adc = reinterpret_cast<unsigned char*>(t_["slot_adc"].data_ptr<uint8_t>());
+      S_.slot_ag = reinterpret_cast<unsigned char*>(t_["slot_ag"].data_ptr<uint8_t>());
+      S_.slot_mdc = reinterpret_cast<unsigned char*>(t_["slot_mdc"].data_ptr<uint8_t>());
+      S_.slot_mg = reinterpret_cast<unsigned char*>(t_["slot_mg"].data_ptr<uint8_t>());
+      S_.slot_has_rl = reinterpret_cast<unsigned char*>(t_["slot_has_rl"].data_ptr<uint8_t>());
+      S_.slot_nrew = reinterpret_cast<unsigned char*>(t_["slot_nrew"].data_ptr<uint8_t>());
+      T_PTR(x_s0, float);
+      S_.x_nsel = reinterpret_cast<short*>(t_["x_nsel"].data_ptr<int16_t>());
+      S_.x_adc = reinterpret_cast<unsigned char*>(t_["x_adc"].data_ptr<uint8_t>());
+      S_.x_ag = reinterpret_cast<unsigned char*>(t_["x_ag"].data_ptr<uint8_t>());
+      S_.x_mdc = reinterpret_cast<unsigned char*>(t_["x_mdc"].data_ptr<uint8_t>());
+      S_.x_mg = reinterpret_cast<unsigned char*>(t_["x_mg"].data_ptr<uint8_t>());
+      S_.x_has_rl = reinterpret_cast<unsigned char*>(t_["x_has_rl"].data_ptr<uint8_t>());
+      T_PTR(lat_hist, int);
+      S_.lat_count = reinterpret_cast<long long*>(t_["lat_count"].data_ptr<int64_t>());
+      T_PTR(lat_sum, double);
+      T_PTR(tr_count, int);
+      T_PTR(tr_s0, float); T_PTR(tr_s1, float);
+      S_.tr_adc = reinterpret_cast<unsigned char*>(t_["tr_adc"].data_ptr<uint8_t>());
+      S_.tr_ag = reinterpret_cast<unsigned char*>(t_["tr_ag"].data_ptr<uint8_t>());
+      T_PTR(tr_r, float); T_PTR(tr_costs, float);
+      S_.tr_mdc = reinterpret_cast<unsigned char*>(t_["tr_mdc"].data_ptr<uint8_t>());
+      S_.tr_mg = reinterpret_cast<unsigned char*>(t_["tr_mg"].data_ptr<uint8_t>());
+    }
   }
 
   // launch one advance chunk; returns immediately (stream-async)
@@ -926,6 +1415,8 @@ class BatchedSimHip {
         hipLaunchKernelGGL(advance_kernel<A_ECO_ROUTE>, grid, block, 0, stream, S_, t_target, max_ev); break;
       case A_DEBUG:
         hipLaunchKernelGGL(advance_kernel<A_DEBUG>, grid, block, 0, stream, S_, t_target, max_ev); break;
+      case A_CHSAC:
+        hipLaunchKernelGGL(advance_kernel<A_CHSAC>, grid, block, 0, stream, S_, t_target, max_ev); break;
       default:
         throw std::runtime_error("unsupported algo for HIP engine");
     }

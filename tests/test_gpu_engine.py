@@ -238,3 +238,55 @@ def test_bench_contract():
         assert k in out, k
     assert out["value"] > 0 and out["n_gpus"] == 1
     assert out["metric"] == "sim_events_per_sec"
+
+
+@needs_gpu
+def test_chsac_batched_runs_and_trains(tmp_path):
+    """CHSAC-AF on the batched engine: replicas pause at decision points, the
+    host serves batched policy actions, transitions flow into replay, SAC
+    trains on-device (BASELINE.json config 4 capability)."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="poisson", rate=2.0)
+    trn = ArrivalProcess(mode="poisson", rate=0.3)
+    out = str(tmp_path / "rl")
+    eng = BatchedEngine(sc, inf, trn, algo="chsac_af", replicas=32,
+                        duration=120.0, log_interval=5.0, out_dir=out,
+                        seed=7, enable_logs=True,
+                        rl_warmup=64, rl_batch=32, rl_train_interval=16,
+                        events_per_launch=5000)
+    st = eng.run()
+    assert st["jobs_completed"] > 0
+    assert eng.replay.size > 0, "no transitions reached the replay ring"
+    assert eng.rl_updates > 0, "SAC never trained"
+    assert int(eng.t["err"].max().item()) == 0
+    # transitions carry sane rewards/costs
+    b = eng.replay.sample(16)
+    assert torch.isfinite(b["r"]).all()
+    assert (b["costs"]["latency_p99"] >= 0).all()
+    assert (b["costs"]["gpu_over"] >= 0).all()
+    # logging replica produced job rows
+    with open(os.path.join(out, "job_log.csv")) as fh:
+        rows = list(csv.DictReader(fh))
+    assert rows
+
+
+@needs_gpu
+def test_chsac_batched_respects_masks():
+    """Actions applied must come from the masked policy: chosen DC always has
+    free GPUs at selection time (mask honored end-to-end)."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="off", rate=0.0)
+    trn = ArrivalProcess(mode="poisson", rate=1.0)
+    eng = BatchedEngine(sc, inf, trn, algo="chsac_af", replicas=16,
+                        duration=60.0, log_interval=5.0, out_dir=None,
+                        seed=3, enable_logs=False, rl_warmup=10**9,
+                        events_per_launch=5000)
+    st = eng.run()
+    assert st["events"] > 0
+    assert int(eng.t["err"].max().item()) == 0
