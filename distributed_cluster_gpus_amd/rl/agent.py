@@ -86,10 +86,16 @@ class CHSACAgent:
         return stats
 
     def enable_ddp(self):
-        """Install the RCCL flat-gradient all-reduce hook (call after
-        torch.distributed.init_process_group; see parallel/dist.py)."""
-        from ..parallel.dist import allreduce_gradients
+        """Install data-parallel hooks (call after
+        torch.distributed.init_process_group; see parallel/dist.py):
+        fused flat gradient all-reduce after each backward, and cross-rank
+        averaging of constraint-cost means so the PID-lambda state stays
+        identical on every replica."""
+        from ..parallel.dist import allreduce_gradients, allreduce_scalar, world_size
         self.algo.allreduce_hook = allreduce_gradients
+        w = world_size()
+        self.cmdp.cost_reduce_hook = (
+            lambda x: allreduce_scalar(x, device=self.device) / w)
 
     # ---- checkpointing (capability extension; the reference has none) ----
     def state_dict(self) -> Dict:

@@ -33,6 +33,9 @@ class PIDLagrangian:
         self.lmbda = {k: torch.tensor(0.0) for k in constraints}
         self.err_int = {k: 0.0 for k in constraints}
         self.err_prev = {k: 0.0 for k in constraints}
+        # data-parallel hook: averages a scalar cost mean across ranks so the
+        # PID state (and hence lambda) stays identical on every DP replica
+        self.cost_reduce_hook = None
 
     def effective_reward(self, r: torch.Tensor,
                          cost_dict: Dict[str, torch.Tensor]) -> torch.Tensor:
@@ -50,6 +53,8 @@ class PIDLagrangian:
             if name not in cost_dict:
                 continue
             c_mean = float(cost_dict[name].mean().item())
+            if self.cost_reduce_hook is not None:
+                c_mean = self.cost_reduce_hook(c_mean)
             e = max(0.0, c_mean - spec.target)
             self.err_int[name] += e
             d = e - self.err_prev[name]
