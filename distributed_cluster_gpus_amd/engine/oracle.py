@@ -305,6 +305,9 @@ class OracleEngine:
             job.rl_state0 = obs
             job.rl_action = {"dc_idx": int(a["dc"]), "g_idx": int(a["g"]), "n": n_sel}
             lnet, bw, cost, transfer_s = self._net_tuple(ing_name, dc_name, job)
+        elif self.sc.router.use_weighted:
+            dc_name = self._weighted_route(ing_name, job)
+            lnet, bw, cost, transfer_s = self._net_tuple(ing_name, dc_name, job)
         else:
             dc_name = self.rng.choice(self._dc_names)
             lnet, bw, cost, transfer_s = self._net_tuple(ing_name, dc_name, job)
@@ -318,6 +321,32 @@ class OracleEngine:
         self._schedule(self.now + arr.next_interarrival(self.now, self.rng),
                        "arrival_inf" if jtype == "inference" else "arrival_trn",
                        {"ing": ing_name})
+
+    def _weighted_route(self, ing_name: str, job: JobState) -> str:
+        """Weighted router score  wE*E1 + wL*Lnet + wC*(E1*CI)  with optional
+        power-of-d candidate sampling.  The reference defines these weights
+        (router.py:3-9) but routes uniformly at random; this is the opt-in
+        implementation (Scenario.router.use_weighted=True) — SURVEY §2 row 10
+        'implement the weighted score for parity-plus'."""
+        rp = self.sc.router
+        if rp.d_choices and 0 < rp.d_choices < len(self._dc_names):
+            cands = self.rng.sample(self._dc_names, rp.d_choices)
+        else:
+            cands = self._dc_names
+        best_name, best_score = None, None
+        for name in cands:
+            dc = self.dcs[name]
+            pC, tC = self._coeffs(name, job.jtype)
+            _, _, _, _, E_unit = best_nf_grid(
+                self.sc.policy.max_gpus_per_job, dc.freq_levels, pC, tC,
+                objective="energy", deadline_s=job.deadline)
+            E1 = E_unit * job.size  # J per job at the energy-optimal point
+            lnet, _, _, _ = self._net_tuple(ing_name, name, job)
+            CI = self.sc.carbon_intensity.get(name, 0.0)
+            score = rp.w_energy * E1 + rp.w_latency * lnet + rp.w_carbon * (E1 * CI)
+            if best_score is None or score < best_score:
+                best_name, best_score = name, score
+        return best_name
 
     def _score_dc(self, dc: DataCenterState, job: JobState):
         """Eco-route DC score: (score, n*, f*) for the configured objective
