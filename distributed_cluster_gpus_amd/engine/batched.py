@@ -54,7 +54,7 @@ class BatchedEngine:
                  logger=None, show_progress: bool = False,
                  device: Optional[torch.device] = None,
                  rank: int = 0, world: int = 1,
-                 tcap: int = 256, qcap: int = 24576,
+                 tcap: int = 64, qcap: int = 24576,
                  events_per_launch: int = 50000,
                  enable_logs: bool = True,
                  sla_p99_ms: float = 500.0, energy_budget_j=None,

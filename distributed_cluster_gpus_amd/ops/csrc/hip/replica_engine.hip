@@ -37,10 +37,6 @@
 
 namespace dcg {
 
-#ifndef DCG_WAVES_PER_EU
-#define DCG_WAVES_PER_EU 4
-#endif
-
 constexpr int MAX_DC = 8;
 constexpr int MAX_FREQ = 8;
 
@@ -493,7 +489,7 @@ __device__ void accrue_to(Ctx& c, double t) {
 }
 
 // emit one cluster-log row set (logging replica only; wave-cooperative counts)
-__device__ __attribute__((noinline)) void emit_cluster_rows(Ctx& c, double now) {
+__device__ void emit_cluster_rows(Ctx& c, double now) {
   const EngineDesc& S = *c.S;
   if (c.r != S.log_replica) return;
   int64_t base = (int64_t)c.r * S.total_slots;
@@ -537,7 +533,7 @@ __device__ __attribute__((noinline)) void emit_cluster_rows(Ctx& c, double now) 
   }
 }
 
-__device__ __attribute__((noinline)) void emit_job_row(Ctx& c, int d, int jt, int jid, int ing,
+__device__ void emit_job_row(Ctx& c, int d, int jt, int jid, int ing,
                              float size, double fused, int n, float netlat,
                              double start, double finish) {
   const EngineDesc& S = *c.S;
@@ -560,7 +556,7 @@ __device__ __attribute__((noinline)) void emit_job_row(Ctx& c, int d, int jt, in
 // cap_greedy controller step at log ticks (reference :248-315, v1: repeatedly
 // apply the currently-cheapest single down-step atom with exact re-estimation;
 // distributionally equivalent to the reference's per-pass sorted ladders)
-__device__ __attribute__((noinline)) void cap_greedy_control(Ctx& c, double now) {
+__device__ void cap_greedy_control(Ctx& c, double now) {
   const EngineDesc& S = *c.S;
   double f_min = S.freq_levels[0];
   for (int k = 1; k < S.n_freq; ++k) f_min = fmin(f_min, S.freq_levels[k]);
@@ -707,7 +703,7 @@ __device__ void rl_build_masks(Ctx& c, int& mdc, int& mg) {
 }
 
 // write an action request and stash the paused-event context
-__device__ __attribute__((noinline)) void rl_request(Ctx& c, int kind, double now, int jt, int ing,
+__device__ void rl_request(Ctx& c, int kind, double now, int jt, int ing,
                            float size, float netlat, int jid, int src_dc,
                            int from_inf) {
   const EngineDesc& S = *c.S;
@@ -802,7 +798,7 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, float size, float netlat,
 }
 
 // emit a CHSAC transition (s0, s1, a, r, costs, masks) into the global ring
-__device__ __attribute__((noinline)) void rl_emit_transition(Ctx& c, const float* s0, int a_dc, int a_g,
+__device__ void rl_emit_transition(Ctx& c, const float* s0, int a_dc, int a_g,
                                    float r, float c_lat, float c_pow,
                                    float c_over, int mdc, int mg, double now) {
   const EngineDesc& S = *c.S;
@@ -837,7 +833,7 @@ __device__ __attribute__((noinline)) void rl_emit_transition(Ctx& c, const float
 
 // ---------------- the advance kernel ----------------
 template <int ALGO>
-__global__ void __launch_bounds__(256, DCG_WAVES_PER_EU)
+__global__ void __launch_bounds__(256)
 advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   int wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   int lane = threadIdx.x & 63;
