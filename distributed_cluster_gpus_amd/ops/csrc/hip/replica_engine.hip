@@ -210,13 +210,40 @@ __device__ __forceinline__ int wave_sum_i32(int v) {
 }
 
 __device__ __forceinline__ void store_fence() {
-  // make lane-0 global stores visible to this wave's subsequent loads
-  __builtin_amdgcn_s_waitcnt(0);
+  // make lane-0 stores (global AND LDS) visible to this wave's subsequent
+  // loads: full hardware wait + compiler reordering barrier
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
 }
+
+// ---------------- hot per-replica state (LDS-resident) ----------------
+// The fields every event touches live in LDS for the kernel's lifetime
+// (~850 B per replica, ~3.4 KB per 4-replica workgroup of the CU's 160 KiB)
+// and are written back to HBM at kernel exit.  This removes most of the
+// dependent global-memory round trips per event (PMC before: waves waiting
+// 63% of cycles on memory; see profiles/README.md).
+constexpr int MAX_STREAMS = 16;
+struct Hot {
+  double arr_next[MAX_STREAMS];
+  double dc_minf[MAX_DC];
+  double p_active[MAX_DC];
+  double sum_tpt[MAX_DC];
+  double energy[MAX_DC];
+  double util_time[MAX_DC];
+  double acc_unit[MAX_DC];
+  double util_begin[MAX_DC];
+  double next_log;
+  float cur_freq[MAX_DC];
+  int dc_mins[MAX_DC];
+  int busy[MAX_DC];
+  int n_running[MAX_DC];
+  int q_len[MAX_DC * 2];
+  int q_head[MAX_DC * 2];
+};
 
 // ---------------- replica context ----------------
 struct Ctx {
   const EngineDesc* S;
+  Hot* hs;      // LDS-resident hot state of this wave's replica
   int r;        // local replica index
   int lane;
   PhiloxState rng;
@@ -226,12 +253,12 @@ struct Ctx {
   __device__ const double* pc3(int d, int jt) const { return &S->pc[(d * 2 + jt) * 3]; }
   __device__ const double* lc3(int d, int jt) const { return &S->lc[(d * 2 + jt) * 3]; }
   __device__ int free_gpus(int d) const {
-    return S->total_gpus[d] - S->busy[r * S->n_dc + d];
+    return S->total_gpus[d] - hs->busy[d];
   }
   __device__ double dc_power(int d) const {
     int idle = free_gpus(d);
     double pi = S->power_gating[d] ? S->p_sleep[d] : S->p_idle[d];
-    return S->p_active[r * S->n_dc + d] + idle * pi;
+    return hs->p_active[d] + idle * pi;
   }
   __device__ double price_kwh(double t) const {
     int h = static_cast<int>(fmod(t, 86400.0) / 3600.0);
@@ -253,8 +280,8 @@ __device__ void rescan_dc_min(Ctx& c, int d) {
   double best = wave_argmin_f64(v, wl);
   slot = __shfl(slot, wl, 64);
   if (c.lane == 0) {
-    S.dc_min_finish[c.r * S.n_dc + d] = best;
-    S.dc_min_slot[c.r * S.n_dc + d] = best < D_INF ? slot : -1;
+    c.hs->dc_minf[d] = best;
+    c.hs->dc_mins[d] = best < D_INF ? slot : -1;
   }
   store_fence();
 }
@@ -289,14 +316,13 @@ __device__ void start_job(Ctx& c, int d, int jt, float size, float netlat,
     S.s_gpus[base + cand] = (short)n;
     S.s_jtype[base + cand] = (char)jt;
     S.s_ing[base + cand] = (char)ing;
-    int rd = c.r * S.n_dc + d;
-    S.busy[rd] += n;
-    S.n_running[rd] += 1;
-    S.p_active[rd] += d_job_power(n, f, c.pc3(d, jt));
-    S.sum_tpt[rd] += 1.0 / T;
-    if (finish < S.dc_min_finish[rd]) {
-      S.dc_min_finish[rd] = finish;
-      S.dc_min_slot[rd] = cand;
+    c.hs->busy[d] += n;
+    c.hs->n_running[d] += 1;
+    c.hs->p_active[d] += d_job_power(n, f, c.pc3(d, jt));
+    c.hs->sum_tpt[d] += 1.0 / T;
+    if (finish < c.hs->dc_minf[d]) {
+      c.hs->dc_minf[d] = finish;
+      c.hs->dc_mins[d] = cand;
     }
   }
   store_fence();
@@ -305,15 +331,14 @@ __device__ void start_job(Ctx& c, int d, int jt, float size, float netlat,
 // heuristic allocator (policy.py:16-41 semantics); mutates cur_freq; returns g
 __device__ int heuristic_alloc(Ctx& c, int d, int jt) {
   const EngineDesc& S = *c.S;
-  int rd = c.r * S.n_dc + d;
   int free = c.free_gpus(d);
   int g = free > 0 ? min(free, S.max_gpj) : 0;
-  double cf = S.cur_freq[rd];
+  double cf = c.hs->cur_freq[d];
   double nf = cf;
   if (!S.energy_aware) {  // perf_first
     if (jt == 0) nf = S.dvfs_high;
     else {
-      int qi = S.q_len[(rd) * 2 + 0];
+      int qi = c.hs->q_len[d * 2 + 0];
       nf = fmax(cf, qi > 0 ? S.dvfs_high : S.default_freq[d]);
     }
   } else {
@@ -323,7 +348,7 @@ __device__ int heuristic_alloc(Ctx& c, int d, int jt) {
       g = min(free, S.max_gpj);
     } else nf = fmax(cf, S.dvfs_low);
   }
-  if (c.lane == 0) S.cur_freq[rd] = (float)nf;
+  if (c.lane == 0) c.hs->cur_freq[d] = (float)nf;
   store_fence();
   return max(1, g);
 }
@@ -372,7 +397,7 @@ __device__ void decide_nf(Ctx& c, int d, int jt, float size, double now,
                                S.n_freq, S.num_fixed);
   } else {  // heuristic family
     n_out = heuristic_alloc(c, d, jt);
-    f_out = S.cur_freq[c.r * S.n_dc + d];
+    f_out = c.hs->cur_freq[d];
   }
 }
 
@@ -380,23 +405,23 @@ __device__ void decide_nf(Ctx& c, int d, int jt, float size, double now,
 __device__ bool queue_push(Ctx& c, int d, int jt, float size, float netlat,
                            int jid, int ing) {
   const EngineDesc& S = *c.S;
-  int q = (c.r * S.n_dc + d) * 2 + jt;
-  int len = S.q_len[q];
+  int ql = d * 2 + jt;
+  int len = c.hs->q_len[ql];
   if (len >= S.qcap) {
     if (c.lane == 0) atomicOr(&S.err[c.r], ERR_QUEUE_OVF);
     return false;
   }
-  int pos = (S.q_head[q] + len) % S.qcap;
+  int pos = (c.hs->q_head[ql] + len) % S.qcap;
   if (c.lane == 0) {
-    int64_t at = (int64_t)q * S.qcap + pos;
+    int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
     S.q_size[at] = size;
     if (c.r == S.log_replica) {
-      int64_t aux = ((int64_t)(d * 2 + jt)) * S.qcap + pos;
+      int64_t aux = ((int64_t)ql) * S.qcap + pos;
       S.q_netlat[aux] = netlat;
       S.q_jid[aux] = jid;
       S.q_ing[aux] = (char)ing;
     }
-    S.q_len[q] = len + 1;
+    c.hs->q_len[ql] = len + 1;
   }
   store_fence();
   return true;
@@ -405,24 +430,24 @@ __device__ bool queue_push(Ctx& c, int d, int jt, float size, float netlat,
 __device__ bool queue_push_front(Ctx& c, int d, int jt, float size,
                                  float netlat, int jid, int ing) {
   const EngineDesc& S = *c.S;
-  int q = (c.r * S.n_dc + d) * 2 + jt;
-  int len = S.q_len[q];
+  int ql = d * 2 + jt;
+  int len = c.hs->q_len[ql];
   if (len >= S.qcap) {
     if (c.lane == 0) atomicOr(&S.err[c.r], ERR_QUEUE_OVF);
     return false;
   }
-  int pos = (S.q_head[q] - 1 + S.qcap) % S.qcap;
+  int pos = (c.hs->q_head[ql] - 1 + S.qcap) % S.qcap;
   if (c.lane == 0) {
-    int64_t at = (int64_t)q * S.qcap + pos;
+    int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
     S.q_size[at] = size;
     if (c.r == S.log_replica) {
-      int64_t aux = ((int64_t)(d * 2 + jt)) * S.qcap + pos;
+      int64_t aux = ((int64_t)ql) * S.qcap + pos;
       S.q_netlat[aux] = netlat;
       S.q_jid[aux] = jid;
       S.q_ing[aux] = (char)ing;
     }
-    S.q_head[q] = pos;
-    S.q_len[q] = len + 1;
+    c.hs->q_head[ql] = pos;
+    c.hs->q_len[ql] = len + 1;
   }
   store_fence();
   return true;
@@ -431,14 +456,14 @@ __device__ bool queue_push_front(Ctx& c, int d, int jt, float size,
 __device__ bool queue_pop(Ctx& c, int d, int jt, float& size, float& netlat,
                           int& jid, int& ing) {
   const EngineDesc& S = *c.S;
-  int q = (c.r * S.n_dc + d) * 2 + jt;
-  int len = S.q_len[q];
+  int ql = d * 2 + jt;
+  int len = c.hs->q_len[ql];
   if (len <= 0) return false;
-  int pos = S.q_head[q];
-  int64_t at = (int64_t)q * S.qcap + pos;
+  int pos = c.hs->q_head[ql];
+  int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
   size = S.q_size[at];
   if (c.r == S.log_replica) {
-    int64_t aux = ((int64_t)(d * 2 + jt)) * S.qcap + pos;
+    int64_t aux = ((int64_t)ql) * S.qcap + pos;
     netlat = S.q_netlat[aux];
     jid = S.q_jid[aux];
     ing = S.q_ing[aux];
@@ -446,8 +471,8 @@ __device__ bool queue_pop(Ctx& c, int d, int jt, float& size, float& netlat,
     netlat = 0.0f; jid = 0; ing = 0;
   }
   if (c.lane == 0) {
-    S.q_head[q] = (pos + 1) % S.qcap;
-    S.q_len[q] = len - 1;
+    c.hs->q_head[ql] = (pos + 1) % S.qcap;
+    c.hs->q_len[ql] = len - 1;
   }
   store_fence();
   return true;
@@ -475,14 +500,13 @@ __device__ void drain_queues(Ctx& c, int d, double now) {
 __device__ void accrue_to(Ctx& c, double t) {
   const EngineDesc& S = *c.S;
   if (c.lane < S.n_dc) {
-    int rd = c.r * S.n_dc + c.lane;
     double last = c.now;
     if (last < 0.0) {
-      S.util_begin[rd] = t;
+      c.hs->util_begin[c.lane] = t;
     } else {
       double dt = fmax(0.0, t - last);
-      S.util_time[rd] += S.busy[rd] * dt;
-      S.energy_j[rd] += c.dc_power(c.lane) * dt;
+      c.hs->util_time[c.lane] += c.hs->busy[c.lane] * dt;
+      c.hs->energy[c.lane] += c.dc_power(c.lane) * dt;
     }
   }
   store_fence();
@@ -499,30 +523,29 @@ __device__ void emit_cluster_rows(Ctx& c, double now) {
     for (int k = lo + c.lane; k < hi; k += 64)
       if (S.s_gpus[base + k] != 0 && S.s_jtype[base + k] == 0) cnt_inf++;
     cnt_inf = wave_sum_i32(cnt_inf);
-    int rd = c.r * S.n_dc + d;
-    int run_total = S.n_running[rd];
+    int run_total = c.hs->n_running[d];
     if (c.lane == 0) {
       int idx = *S.cl_count;
       if (idx < S.cl_cap) {
         double* row = &S.cl_rows[(int64_t)idx * 16];
-        double begin = S.util_begin[rd];
+        double begin = c.hs->util_begin[d];
         double elapsed = fmax(1e-9, now - (begin >= 0 ? begin : now));
         row[0] = now;
         row[1] = d;
-        row[2] = S.cur_freq[rd];
-        row[3] = S.busy[rd];
-        row[4] = S.total_gpus[d] - S.busy[rd];
+        row[2] = c.hs->cur_freq[d];
+        row[3] = c.hs->busy[d];
+        row[4] = S.total_gpus[d] - c.hs->busy[d];
         row[5] = run_total;
         row[6] = cnt_inf;
         row[7] = run_total - cnt_inf;
-        row[8] = S.q_len[rd * 2 + 0];
-        row[9] = S.q_len[rd * 2 + 1];
-        row[10] = S.total_gpus[d] ? (double)S.busy[rd] / S.total_gpus[d] : 0.0;
+        row[8] = c.hs->q_len[d * 2 + 0];
+        row[9] = c.hs->q_len[d * 2 + 1];
+        row[10] = S.total_gpus[d] ? (double)c.hs->busy[d] / S.total_gpus[d] : 0.0;
         row[11] = S.total_gpus[d]
-            ? S.util_time[rd] / (S.total_gpus[d] * elapsed) : 0.0;
-        row[12] = S.acc_unit[rd];
+            ? c.hs->util_time[d] / (S.total_gpus[d] * elapsed) : 0.0;
+        row[12] = c.hs->acc_unit[d];
         row[13] = c.dc_power(d);
-        row[14] = S.energy_j[rd];
+        row[14] = c.hs->energy[d];
         row[15] = 0.0;
         *S.cl_count = idx + 1;
       } else {
@@ -612,10 +635,9 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
     double remaining_units = fmax(0.0, (finish_old - now)) / fmax(T_old, 1e-12);
     double finish_new = now + remaining_units * T_new;
     if (c.lane == 0) {
-      int rd = c.r * S.n_dc + d;
-      S.p_active[rd] += d_job_power(n, best_fto, c.pc3(d, jt)) -
-                        d_job_power(n, old_f, c.pc3(d, jt));
-      S.sum_tpt[rd] += 1.0 / T_new - 1.0 / T_old;
+      c.hs->p_active[d] += d_job_power(n, best_fto, c.pc3(d, jt)) -
+                           d_job_power(n, old_f, c.pc3(d, jt));
+      c.hs->sum_tpt[d] += 1.0 / T_new - 1.0 / T_old;
       S.s_fused[base + best_slot] = (float)best_fto;
       S.s_finish[base + best_slot] = finish_new;
     }
@@ -632,15 +654,14 @@ __device__ void rl_build_obs(Ctx& c, double now, float* out) {
   const EngineDesc& S = *c.S;
   if (c.lane == 0) out[0] = (float)now;
   if (c.lane < S.n_dc) {
-    int rd = c.r * S.n_dc + c.lane;
     float total = (float)S.total_gpus[c.lane];
-    float busy = (float)S.busy[rd];
+    float busy = (float)c.hs->busy[c.lane];
     out[1 + 6 * c.lane + 0] = total;
     out[1 + 6 * c.lane + 1] = busy;
     out[1 + 6 * c.lane + 2] = fmaxf(0.0f, total - busy);
-    out[1 + 6 * c.lane + 3] = S.cur_freq[rd];
-    out[1 + 6 * c.lane + 4] = (float)S.q_len[rd * 2 + 0];
-    out[1 + 6 * c.lane + 5] = (float)S.q_len[rd * 2 + 1];
+    out[1 + 6 * c.lane + 3] = c.hs->cur_freq[c.lane];
+    out[1 + 6 * c.lane + 4] = (float)c.hs->q_len[c.lane * 2 + 0];
+    out[1 + 6 * c.lane + 5] = (float)c.hs->q_len[c.lane * 2 + 1];
   }
   store_fence();
 }
@@ -784,14 +805,13 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, float size, float netlat,
     S.slot_mg[base + cand] = (unsigned char)mg;
     S.slot_has_rl[base + cand] = 1;
     S.slot_nrew[base + cand] = (unsigned char)max(1, n_rew);
-    int rd = c.r * S.n_dc + d;
-    S.busy[rd] += n;
-    S.n_running[rd] += 1;
-    S.p_active[rd] += d_job_power(n, f, c.pc3(d, jt));
-    S.sum_tpt[rd] += 1.0 / T;
-    if (finish < S.dc_min_finish[rd]) {
-      S.dc_min_finish[rd] = finish;
-      S.dc_min_slot[rd] = cand;
+    c.hs->busy[d] += n;
+    c.hs->n_running[d] += 1;
+    c.hs->p_active[d] += d_job_power(n, f, c.pc3(d, jt));
+    c.hs->sum_tpt[d] += 1.0 / T;
+    if (finish < c.hs->dc_minf[d]) {
+      c.hs->dc_minf[d] = finish;
+      c.hs->dc_mins[d] = cand;
     }
   }
   store_fence();
@@ -844,6 +864,11 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   c.r = wave;
   c.lane = lane;
   if (S.done[c.r]) return;
+  // CHSAC fast-path: a replica still waiting for its policy response must
+  // not advance (and must not round-trip the hot state)
+  if (ALGO == A_CHSAC && S.pend_kind[c.r] != PEND_NONE &&
+      S.req_flag[c.r] != REQ_READY)
+    return;
   c.now = S.now[c.r];
   c.rng.key = S.seed ^ (0x9E3779B97F4A7C15ull * (uint64_t)(S.rep_id_offset + c.r));
   c.rng.ctr = S.rng_ctr[c.r];
@@ -853,12 +878,37 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   long long n_events = 0;
   bool paused = false;
 
+  // ---- load the hot per-replica state into LDS (lane-parallel) ----
+  __shared__ Hot hot_block[4];
+  c.hs = &hot_block[threadIdx.x >> 6];
+  {
+    Hot* h = c.hs;
+    int nd = S.n_dc;
+    if (lane < NS) h->arr_next[lane] = S.arr_next[(int64_t)c.r * NS + lane];
+    if (lane < nd) {
+      int rd = c.r * nd + lane;
+      h->dc_minf[lane] = S.dc_min_finish[rd];
+      h->dc_mins[lane] = S.dc_min_slot[rd];
+      h->p_active[lane] = S.p_active[rd];
+      h->sum_tpt[lane] = S.sum_tpt[rd];
+      h->energy[lane] = S.energy_j[rd];
+      h->util_time[lane] = S.util_time[rd];
+      h->acc_unit[lane] = S.acc_unit[rd];
+      h->util_begin[lane] = S.util_begin[rd];
+      h->cur_freq[lane] = S.cur_freq[rd];
+      h->busy[lane] = S.busy[rd];
+      h->n_running[lane] = S.n_running[rd];
+    }
+    if (lane < nd * 2) {
+      h->q_len[lane] = S.q_len[c.r * nd * 2 + lane];
+      h->q_head[lane] = S.q_head[c.r * nd * 2 + lane];
+    }
+    if (lane == 0) h->next_log = S.next_log[c.r];
+  }
+  store_fence();
+
   // ---- CHSAC: resume a paused action request ----
   if (ALGO == A_CHSAC && S.pend_kind[c.r] != PEND_NONE) {
-    if (S.req_flag[c.r] != REQ_READY) {
-      if (lane == 0) S.rng_ctr[c.r] = c.rng.ctr;
-      return;  // still waiting for the host policy
-    }
     int a_dc = S.resp_dc[c.r];
     int a_g = S.resp_g[c.r];
     const float* s0 = &S.req_obs[(int64_t)c.r * S.obs_dim];
@@ -934,18 +984,18 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     int kind = -1, idx = -1;
     // arrival streams on lanes [0, NS)
     if (lane < NS) {
-      double t = S.arr_next[(int64_t)c.r * NS + lane];
+      double t = c.hs->arr_next[lane];
       if (t < v) { v = t; kind = 0; idx = lane; }
     }
     // log tick on lane NS
     if (lane == NS) {
-      double t = S.next_log[c.r];
+      double t = c.hs->next_log;
       if (t < v) { v = t; kind = 3; idx = 0; }
     }
     // dc min finishes on lanes [32, 32+n_dc)
     if (lane >= 32 && lane < 32 + S.n_dc) {
-      double t = S.dc_min_finish[c.r * S.n_dc + (lane - 32)];
-      if (t < v) { v = t; kind = 2; idx = S.dc_min_slot[c.r * S.n_dc + (lane - 32)]; }
+      double t = c.hs->dc_minf[lane - 32];
+      if (t < v) { v = t; kind = 2; idx = c.hs->dc_mins[lane - 32]; }
     }
     // transfers: strided over tcap
     for (int k = lane; k < S.tcap; k += 64) {
@@ -960,19 +1010,18 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     if (t_min > S.end_time) {
       // ---- end of simulation for this replica: final flush ----
       if (lane < S.n_dc) {
-        int rd = c.r * S.n_dc + lane;
         double last = c.now;
         if (last >= 0.0 && last < S.end_time) {
-          S.util_time[rd] += S.busy[rd] * (S.end_time - last);
+          c.hs->util_time[lane] += c.hs->busy[lane] * (S.end_time - last);
           // reference quirk: the final accrue_energy(end) uses the BASELINE
           // idle/sleep + p_peak*f^alpha model (models.py:82-91)
-          double f = S.cur_freq[rd];
-          int active = S.busy[rd];
+          double f = c.hs->cur_freq[lane];
+          int active = c.hs->busy[lane];
           int idlec = S.total_gpus[lane] - active;
           double pa = active * (S.p_idle[lane] +
                                 S.p_peak[lane] * pow(f, S.pow_alpha[lane]));
           double pi = idlec * (S.power_gating[lane] ? S.p_sleep[lane] : S.p_idle[lane]);
-          S.energy_j[rd] += (pa + pi) * (S.end_time - last);
+          c.hs->energy[lane] += (pa + pi) * (S.end_time - last);
         }
       }
       if (lane == 0) S.done[c.r] = 1;
@@ -1018,7 +1067,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
             }
           }
         }
-        if (lane == 0) S.arr_next[(int64_t)c.r * NS + idx] = t_min + ia_rl;
+        if (lane == 0) c.hs->arr_next[idx] = t_min + ia_rl;
         store_fence();
         rl_request(c, PEND_ARRIVAL, t_min, jt, ing, (float)size, 0.0f, jid,
                    -1, 0);
@@ -1102,7 +1151,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           }
         }
       }
-      if (lane == 0) S.arr_next[(int64_t)c.r * NS + idx] = t_min + ia;
+      if (lane == 0) c.hs->arr_next[idx] = t_min + ia;
       store_fence();
 
     } else if (kind == 1) {
@@ -1151,15 +1200,14 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int ingr = S.s_ing[at];
       double T = d_unit_time(n, fused, c.lc3(d, jt));
       if (lane == 0) {
-        int rd = c.r * S.n_dc + d;
         S.s_finish[at] = D_INF;
         S.s_gpus[at] = 0;
-        S.busy[rd] = max(0, S.busy[rd] - n);
-        S.n_running[rd] -= 1;
-        S.p_active[rd] -= d_job_power(n, fused, c.pc3(d, jt));
-        S.sum_tpt[rd] -= 1.0 / T;
+        c.hs->busy[d] = max(0, c.hs->busy[d] - n);
+        c.hs->n_running[d] -= 1;
+        c.hs->p_active[d] -= d_job_power(n, fused, c.pc3(d, jt));
+        c.hs->sum_tpt[d] -= 1.0 / T;
         // remainder job-units: window = finish mod log_interval (quirk)
-        S.acc_unit[rd] += (1.0 / T) * fmod(t_min, S.log_interval);
+        c.hs->acc_unit[d] += (1.0 / T) * fmod(t_min, S.log_interval);
         // metrics
         S.jobs_done[c.r] += 1;
         S.sum_lat[c.r] += t_min - start;
@@ -1240,11 +1288,10 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           cap_greedy_control(c, t_min);
         } else if (ALGO == A_ECO_ROUTE || ALGO == A_CARBON_COST) {
           if (lane < S.n_dc) {
-            int rd = c.r * S.n_dc + lane;
-            if (S.busy[rd] == 0) {
+            if (c.hs->busy[lane] == 0) {
               double fm = S.freq_levels[0];
               for (int k = 1; k < S.n_freq; ++k) fm = fmin(fm, S.freq_levels[k]);
-              S.cur_freq[rd] = (float)fm;
+              c.hs->cur_freq[lane] = (float)fm;
             }
           }
           store_fence();
@@ -1253,17 +1300,41 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       }
       // acc_job_unit for running jobs: cached sum_tpt * interval
       if (lane < S.n_dc) {
-        int rd = c.r * S.n_dc + lane;
-        S.acc_unit[rd] += S.sum_tpt[rd] * S.log_interval;
+        c.hs->acc_unit[lane] += c.hs->sum_tpt[lane] * S.log_interval;
       }
       store_fence();
       emit_cluster_rows(c, t_min);
-      if (lane == 0) S.next_log[c.r] = t_min + S.log_interval;
+      if (lane == 0) c.hs->next_log = t_min + S.log_interval;
       store_fence();
     }
   }
 
+  // ---- write the hot state back to HBM ----
+  {
+    Hot* h = c.hs;
+    int nd = S.n_dc;
+    if (lane < NS) S.arr_next[(int64_t)c.r * NS + lane] = h->arr_next[lane];
+    if (lane < nd) {
+      int rd = c.r * nd + lane;
+      S.dc_min_finish[rd] = h->dc_minf[lane];
+      S.dc_min_slot[rd] = h->dc_mins[lane];
+      S.p_active[rd] = h->p_active[lane];
+      S.sum_tpt[rd] = h->sum_tpt[lane];
+      S.energy_j[rd] = h->energy[lane];
+      S.util_time[rd] = h->util_time[lane];
+      S.acc_unit[rd] = h->acc_unit[lane];
+      S.util_begin[rd] = h->util_begin[lane];
+      S.cur_freq[rd] = h->cur_freq[lane];
+      S.busy[rd] = h->busy[lane];
+      S.n_running[rd] = h->n_running[lane];
+    }
+    if (lane < nd * 2) {
+      S.q_len[c.r * nd * 2 + lane] = h->q_len[lane];
+      S.q_head[c.r * nd * 2 + lane] = h->q_head[lane];
+    }
+  }
   if (lane == 0) {
+    S.next_log[c.r] = c.hs->next_log;
     S.now[c.r] = c.now;
     S.rng_ctr[c.r] = c.rng.ctr;
     S.ev_count[c.r] += n_events;
