@@ -222,6 +222,7 @@ __device__ __forceinline__ void store_fence() {
 // dependent global-memory round trips per event (PMC before: waves waiting
 // 63% of cycles on memory; see profiles/README.md).
 constexpr int MAX_STREAMS = 16;
+constexpr int WAVES_PER_BLOCK = 2;
 struct Hot {
   double arr_next[MAX_STREAMS];
   double dc_minf[MAX_DC];
@@ -244,6 +245,8 @@ struct Hot {
 struct Ctx {
   const EngineDesc* S;
   Hot* hs;      // LDS-resident hot state of this wave's replica
+  double* l_fin;  // LDS mirror of this replica's s_finish[total_slots]
+  double* l_xt;   // LDS mirror of this replica's x_time[tcap]
   int r;        // local replica index
   int lane;
   PhiloxState rng;
@@ -273,7 +276,7 @@ __device__ void rescan_dc_min(Ctx& c, int d) {
   double v = D_INF;
   int slot = -1;
   for (int k = lo + c.lane; k < hi; k += 64) {
-    double f = S.s_finish[(int64_t)c.r * S.total_slots + k];
+    double f = c.l_fin[k];
     if (f < v) { v = f; slot = k; }
   }
   int wl;
@@ -292,10 +295,10 @@ __device__ void start_job(Ctx& c, int d, int jt, float size, float netlat,
   const EngineDesc& S = *c.S;
   int64_t base = (int64_t)c.r * S.total_slots;
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
-  // find first empty slot (s_gpus == 0)
+  // find first empty slot (finish == INF <=> s_gpus == 0), from the LDS mirror
   int cand = INT_MAX;
   for (int k = lo + c.lane; k < hi; k += 64) {
-    if (S.s_gpus[base + k] == 0) { cand = k; break; }
+    if (c.l_fin[k] >= D_INF) { cand = k; break; }
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1)
@@ -307,7 +310,7 @@ __device__ void start_job(Ctx& c, int d, int jt, float size, float netlat,
   double T = d_unit_time(n, f, c.lc3(d, jt));
   double finish = now + (double)size * T;
   if (c.lane == 0) {
-    S.s_finish[base + cand] = finish;
+    c.l_fin[cand] = finish;
     S.s_start[base + cand] = now;
     S.s_size[base + cand] = size;
     S.s_fused[base + cand] = (float)f;
@@ -595,7 +598,7 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
     int best_slot = -1;
     double best_fto = 0;
     for (int k = c.lane; k < S.total_slots; k += 64) {
-      if (S.s_gpus[base + k] == 0) continue;
+      if (c.l_fin[k] >= D_INF) continue;
       int d = S.slot_dc[k];
       double fu = S.s_fused[base + k];
       if (fu <= f_min + 1e-12) continue;
@@ -631,7 +634,7 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
     double old_f = S.s_fused[base + best_slot];
     double T_old = d_unit_time(n, old_f, c.lc3(d, jt));
     double T_new = d_unit_time(n, best_fto, c.lc3(d, jt));
-    double finish_old = S.s_finish[base + best_slot];
+    double finish_old = c.l_fin[best_slot];
     double remaining_units = fmax(0.0, (finish_old - now)) / fmax(T_old, 1e-12);
     double finish_new = now + remaining_units * T_new;
     if (c.lane == 0) {
@@ -639,7 +642,7 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
                            d_job_power(n, old_f, c.pc3(d, jt));
       c.hs->sum_tpt[d] += 1.0 / T_new - 1.0 / T_old;
       S.s_fused[base + best_slot] = (float)best_fto;
-      S.s_finish[base + best_slot] = finish_new;
+      c.l_fin[best_slot] = finish_new;
     }
     store_fence();
     rescan_dc_min(c, d);
@@ -775,7 +778,7 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, float size, float netlat,
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
   int cand = INT_MAX;
   for (int k = lo + c.lane; k < hi; k += 64) {
-    if (S.s_gpus[base + k] == 0) { cand = k; break; }
+    if (c.l_fin[k] >= D_INF) { cand = k; break; }
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1)
@@ -790,7 +793,7 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, float size, float netlat,
   for (int k = c.lane; k < S.obs_dim; k += 64)
     S.slot_s0[(base + cand) * S.obs_dim + k] = s0[k];
   if (c.lane == 0) {
-    S.s_finish[base + cand] = finish;
+    c.l_fin[cand] = finish;
     S.s_start[base + cand] = now;
     S.s_size[base + cand] = size;
     S.s_fused[base + cand] = (float)f;
@@ -853,7 +856,7 @@ __device__ void rl_emit_transition(Ctx& c, const float* s0, int a_dc, int a_g,
 
 // ---------------- the advance kernel ----------------
 template <int ALGO>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(64 * WAVES_PER_BLOCK)
 advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   int wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   int lane = threadIdx.x & 63;
@@ -878,12 +881,27 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   long long n_events = 0;
   bool paused = false;
 
-  // ---- load the hot per-replica state into LDS (lane-parallel) ----
-  __shared__ Hot hot_block[4];
-  c.hs = &hot_block[threadIdx.x >> 6];
+  // ---- carve the dynamic-LDS region: per wave, Hot + s_finish + x_time ----
+  // (single extern __shared__ object, every carve 16-byte aligned)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  {
+    size_t hot_sz = (sizeof(Hot) + 15) & ~size_t(15);
+    size_t fin_sz = (size_t)S.total_slots * sizeof(double);
+    size_t xt_sz = (size_t)S.tcap * sizeof(double);
+    size_t stride = hot_sz + fin_sz + xt_sz;
+    char* base = smem + (threadIdx.x >> 6) * stride;
+    c.hs = reinterpret_cast<Hot*>(base);
+    c.l_fin = reinterpret_cast<double*>(base + hot_sz);
+    c.l_xt = reinterpret_cast<double*>(base + hot_sz + fin_sz);
+  }
   {
     Hot* h = c.hs;
     int nd = S.n_dc;
+    // slot finish times + transfer times into LDS (lane-strided)
+    for (int k = lane; k < S.total_slots; k += 64)
+      c.l_fin[k] = S.s_finish[sbase + k];
+    for (int k = lane; k < S.tcap; k += 64)
+      c.l_xt[k] = S.x_time[(int64_t)c.r * S.tcap + k];
     if (lane < NS) h->arr_next[lane] = S.arr_next[(int64_t)c.r * NS + lane];
     if (lane < nd) {
       int rd = c.r * nd + lane;
@@ -928,7 +946,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       double xfer = bw > 0.0 ? S.payload_gb[jt] / bw : 0.0;
       int cand = INT_MAX;
       for (int k = lane; k < S.tcap; k += 64) {
-        if (S.x_time[(int64_t)c.r * S.tcap + k] >= D_INF) { cand = k; break; }
+        if (c.l_xt[k] >= D_INF) { cand = k; break; }
       }
 #pragma unroll
       for (int off = 32; off > 0; off >>= 1)
@@ -940,7 +958,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         for (int k = lane; k < S.obs_dim; k += 64)
           S.x_s0[at * S.obs_dim + k] = s0[k];
         if (lane == 0) {
-          S.x_time[at] = c.now + lnet + xfer;
+          c.l_xt[cand] = c.now + lnet + xfer;
           S.x_size[at] = size;
           S.x_netlat[at] = (float)lnet;
           S.x_jid[at] = jid;
@@ -997,9 +1015,9 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       double t = c.hs->dc_minf[lane - 32];
       if (t < v) { v = t; kind = 2; idx = c.hs->dc_mins[lane - 32]; }
     }
-    // transfers: strided over tcap
+    // transfers: strided over tcap (LDS mirror)
     for (int k = lane; k < S.tcap; k += 64) {
-      double t = S.x_time[(int64_t)c.r * S.tcap + k];
+      double t = c.l_xt[k];
       if (t < v) { v = t; kind = 1; idx = k; }
     }
     int wl;
@@ -1111,7 +1129,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       // push transfer record
       int cand = INT_MAX;
       for (int k = lane; k < S.tcap; k += 64) {
-        if (S.x_time[(int64_t)c.r * S.tcap + k] >= D_INF) { cand = k; break; }
+        if (c.l_xt[k] >= D_INF) { cand = k; break; }
       }
 #pragma unroll
       for (int off = 32; off > 0; off >>= 1)
@@ -1120,7 +1138,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         if (lane == 0) atomicOr(&S.err[c.r], ERR_XFER_OVF);
       } else if (lane == 0) {
         int64_t at = (int64_t)c.r * S.tcap + cand;
-        S.x_time[at] = t_min + lnet + xfer;
+        c.l_xt[cand] = t_min + lnet + xfer;
         S.x_size[at] = (float)size;
         S.x_netlat[at] = (float)lnet;
         S.x_jid[at] = jid;
@@ -1163,7 +1181,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       float netlat = S.x_netlat[at];
       int jid = S.x_jid[at];
       int ing = S.x_ing[at];
-      if (lane == 0) S.x_time[at] = D_INF;
+      if (lane == 0) c.l_xt[idx] = D_INF;
       store_fence();
       if (c.free_gpus(d) > 0) {
         if (ALGO == A_CHSAC && S.x_has_rl[at]) {
@@ -1200,7 +1218,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int ingr = S.s_ing[at];
       double T = d_unit_time(n, fused, c.lc3(d, jt));
       if (lane == 0) {
-        S.s_finish[at] = D_INF;
+        c.l_fin[slot] = D_INF;
         S.s_gpus[at] = 0;
         c.hs->busy[d] = max(0, c.hs->busy[d] - n);
         c.hs->n_running[d] -= 1;
@@ -1313,6 +1331,10 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   {
     Hot* h = c.hs;
     int nd = S.n_dc;
+    for (int k = lane; k < S.total_slots; k += 64)
+      S.s_finish[sbase + k] = c.l_fin[k];
+    for (int k = lane; k < S.tcap; k += 64)
+      S.x_time[(int64_t)c.r * S.tcap + k] = c.l_xt[k];
     if (lane < NS) S.arr_next[(int64_t)c.r * NS + lane] = h->arr_next[lane];
     if (lane < nd) {
       int rd = c.r * nd + lane;
@@ -1465,29 +1487,38 @@ class BatchedSimHip {
 
   // launch one advance chunk; returns immediately (stream-async)
   void advance(double t_target, int64_t max_ev) {
-    int waves_per_block = 4;  // 256 threads
+    int waves_per_block = WAVES_PER_BLOCK;
     int blocks = (S_.n_rep + waves_per_block - 1) / waves_per_block;
     dim3 grid(blocks), block(64 * waves_per_block);
+    // dynamic LDS: per wave, Hot + s_finish mirror + x_time mirror
+    size_t hot_sz = (sizeof(Hot) + 15) & ~size_t(15);
+    size_t shmem = waves_per_block *
+        (hot_sz + (size_t)S_.total_slots * sizeof(double) +
+         (size_t)S_.tcap * sizeof(double));
+    if (shmem > 64 * 1024)
+      throw std::runtime_error(
+          "scenario too large for the LDS-mirrored engine (total_slots + "
+          "tcap exceed the 64 KiB dynamic-LDS budget per block)");
     hipStream_t stream = at::hip::getCurrentHIPStream();
     switch (S_.algo) {
       case A_DEFAULT:
-        hipLaunchKernelGGL(advance_kernel<A_DEFAULT>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_DEFAULT>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       case A_CAP_UNIFORM:
-        hipLaunchKernelGGL(advance_kernel<A_CAP_UNIFORM>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_CAP_UNIFORM>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       case A_CAP_GREEDY:
-        hipLaunchKernelGGL(advance_kernel<A_CAP_GREEDY>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_CAP_GREEDY>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       case A_JOINT_NF:
-        hipLaunchKernelGGL(advance_kernel<A_JOINT_NF>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_JOINT_NF>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       case A_BANDIT:
-        hipLaunchKernelGGL(advance_kernel<A_BANDIT>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_BANDIT>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       case A_CARBON_COST:
-        hipLaunchKernelGGL(advance_kernel<A_CARBON_COST>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_CARBON_COST>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       case A_ECO_ROUTE:
-        hipLaunchKernelGGL(advance_kernel<A_ECO_ROUTE>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_ECO_ROUTE>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       case A_DEBUG:
-        hipLaunchKernelGGL(advance_kernel<A_DEBUG>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_DEBUG>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       case A_CHSAC:
-        hipLaunchKernelGGL(advance_kernel<A_CHSAC>, grid, block, 0, stream, S_, t_target, max_ev); break;
+        hipLaunchKernelGGL(advance_kernel<A_CHSAC>, grid, block, shmem, stream, S_, t_target, max_ev); break;
       default:
         throw std::runtime_error("unsupported algo for HIP engine");
     }
