@@ -429,7 +429,9 @@ class BatchedEngine:
                 steps = self._tr_since_train // self._rl_train_interval
                 steps = min(steps, 64)  # bound per-launch training work
                 for _ in range(steps):
-                    self.rl.train_step(self.replay.sample(self._rl_batch))
+                    # sync-free SAC step (no stats, tensorized PID)
+                    self.rl.train_step(self.replay.sample(self._rl_batch),
+                                       compute_stats=False)
                     self.rl_updates += 1
                 if steps:
                     self._tr_since_train = 0

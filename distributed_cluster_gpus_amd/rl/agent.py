@@ -49,7 +49,8 @@ class CHSACAgent:
                                     actor_lr=cfg.actor_lr, critic_lr=cfg.critic_lr,
                                     device=self.device)
         self.cmdp = PIDLagrangian(
-            {k: Constraint(name=k, target=v) for k, v in cfg.constraints.items()})
+            {k: Constraint(name=k, target=v) for k, v in cfg.constraints.items()},
+            device=cfg.device)
 
     # ---- acting ----
     def select_action(self, obs: np.ndarray, mask_dc: Optional[np.ndarray],
@@ -72,7 +73,8 @@ class CHSACAgent:
         return None if spec is None else spec.target
 
     # ---- training ----
-    def train_step(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
+    def train_step(self, batch: Dict[str, torch.Tensor],
+                   compute_stats: bool = True) -> Dict[str, float]:
         r = batch["r"]
         costs = batch.get("costs", {})
         r_eff = r.clone()
@@ -81,8 +83,12 @@ class CHSACAgent:
             spec = self.cmdp.constraints[k]
             e = (costs[k] - spec.target).clamp(min=0.0)
             r_eff = r_eff - self.cmdp.lmbda[k].to(r.device) * e
-        stats = self.algo.update({**batch, "r_eff": r_eff})
-        stats.update(self.cmdp.update({k: costs[k] for k in active}))
+        stats = self.algo.update({**batch, "r_eff": r_eff},
+                                 compute_stats=compute_stats)
+        if compute_stats:
+            stats.update(self.cmdp.update({k: costs[k] for k in active}))
+        else:
+            self.cmdp.update_({k: costs[k] for k in active})
         return stats
 
     def enable_ddp(self):
@@ -94,8 +100,9 @@ class CHSACAgent:
         from ..parallel.dist import allreduce_gradients, allreduce_scalar, world_size
         self.algo.allreduce_hook = allreduce_gradients
         w = world_size()
+        from ..parallel.dist import allreduce_tensor_sum
         self.cmdp.cost_reduce_hook = (
-            lambda x: allreduce_scalar(x, device=self.device) / w)
+            lambda t: allreduce_tensor_sum(t.clone()) / w)
 
     # ---- checkpointing (capability extension; the reference has none) ----
     def state_dict(self) -> Dict:

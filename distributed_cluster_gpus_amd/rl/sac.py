@@ -90,7 +90,8 @@ class MaskedHybridSAC(nn.Module):
         return q_min, logp_n
 
     def update(self, batch: Dict[str, torch.Tensor], gamma: float = 0.99,
-               tau: float = 0.005, target_entropy: float = -3.0):
+               tau: float = 0.005, target_entropy: float = -3.0,
+               compute_stats: bool = True):
         dev = self.device
         s = batch["s"].to(dev)
         s_next = batch["s_next"].to(dev)
@@ -148,5 +149,7 @@ class MaskedHybridSAC(nn.Module):
             for tp, p in zip(self.target_critic.parameters(), self.critic.parameters()):
                 tp.data.mul_(1 - tau).add_(tau * p.data)
 
+        if not compute_stats:
+            return {}
         return {"loss_critic": loss_critic.item(), "loss_actor": actor_loss.item(),
                 "loss_temp": temp_loss.item(), "alpha": self.alpha.item()}
