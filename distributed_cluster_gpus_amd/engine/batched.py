@@ -263,7 +263,10 @@ class BatchedEngine:
                 self.rl = make_agent(CHSACAgentConfig(
                     obs_dim=obs_dim, n_dc=n_dc,
                     n_g_choices=int(scenario.policy.max_gpus_per_job),
-                    constraints=constraints, device=str(dev)))
+                    constraints=constraints, device=str(dev),
+                    graph_capturable=(world == 1)))
+            self._graphed = None
+            self._use_graph = (world == 1 and rl_agent is None)
             self.replay = ReplayRing(capacity=int(rl_buffer), obs_dim=obs_dim,
                                      n_costs=3, cost_names=self._cost_names,
                                      n_dc=n_dc,
@@ -428,10 +431,17 @@ class BatchedEngine:
             if self.replay.size >= self._rl_warmup:
                 steps = self._tr_since_train // self._rl_train_interval
                 steps = min(steps, 64)  # bound per-launch training work
+                if steps and self._use_graph and self._graphed is None:
+                    from ..rl.graphed import GraphedSACStep
+                    self._graphed = GraphedSACStep(self.rl, self.replay,
+                                                   self._rl_batch)
                 for _ in range(steps):
-                    # sync-free SAC step (no stats, tensorized PID)
-                    self.rl.train_step(self.replay.sample(self._rl_batch),
-                                       compute_stats=False)
+                    if self._graphed is not None:
+                        self._graphed.step()  # one hipGraph replay
+                    else:
+                        # sync-free eager SAC step (no stats, tensorized PID)
+                        self.rl.train_step(self.replay.sample(self._rl_batch),
+                                           compute_stats=False)
                     self.rl_updates += 1
                 if steps:
                     self._tr_since_train = 0

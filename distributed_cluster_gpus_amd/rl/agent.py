@@ -34,6 +34,7 @@ class CHSACAgentConfig:
     alpha: float = 0.2
     actor_lr: float = 3e-4
     critic_lr: float = 3e-4
+    graph_capturable: bool = False  # Adam capturable state (hipGraph path)
 
 
 class CHSACAgent:
@@ -47,7 +48,8 @@ class CHSACAgent:
         self.algo = MaskedHybridSAC(self.encoder, self.actor, self.critic,
                                     n_quantiles=cfg.n_quantiles, alpha=cfg.alpha,
                                     actor_lr=cfg.actor_lr, critic_lr=cfg.critic_lr,
-                                    device=self.device)
+                                    device=self.device,
+                                    capturable=cfg.graph_capturable)
         self.cmdp = PIDLagrangian(
             {k: Constraint(name=k, target=v) for k, v in cfg.constraints.items()},
             device=cfg.device)

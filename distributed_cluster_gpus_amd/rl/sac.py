@@ -37,7 +37,7 @@ class MaskedHybridSAC(nn.Module):
     def __init__(self, encoder: StateEncoder, actor: HybridActor,
                  critic: TwinQuantileCritic, n_quantiles: int = 32,
                  alpha: float = 0.2, actor_lr: float = 3e-4, critic_lr: float = 3e-4,
-                 device: Optional[torch.device] = None):
+                 device: Optional[torch.device] = None, capturable: bool = False):
         super().__init__()
         self.encoder, self.actor, self.critic = encoder, actor, critic
         self.target_critic = TwinQuantileCritic(encoder.out_dim, actor.n_dc,
@@ -46,12 +46,14 @@ class MaskedHybridSAC(nn.Module):
         self.nq = n_quantiles
         self.taus = torch.linspace(1.0 / (2 * n_quantiles),
                                    1 - 1.0 / (2 * n_quantiles), n_quantiles)
-        self.log_alpha = torch.tensor(math.log(alpha), requires_grad=True)
+        self.device = device or torch.device("cpu")
+        self.log_alpha = torch.tensor(math.log(alpha), requires_grad=True,
+                                      device=self.device)
         self.actor_opt = torch.optim.Adam(
             list(self.encoder.parameters()) + list(self.actor.parameters()) + [self.log_alpha],
-            lr=actor_lr)
-        self.critic_opt = torch.optim.Adam(self.critic.parameters(), lr=critic_lr)
-        self.device = device or torch.device("cpu")
+            lr=actor_lr, capturable=capturable)
+        self.critic_opt = torch.optim.Adam(self.critic.parameters(), lr=critic_lr,
+                                           capturable=capturable)
         self.to(self.device)
         # optional distributed-gradient hook, called after each backward with
         # the parameter list whose grads must be averaged across ranks.
