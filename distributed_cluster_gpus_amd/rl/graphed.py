@@ -36,6 +36,7 @@ class GraphedSACStep:
         self._cpu_gen = torch.Generator().manual_seed(
             int(torch.initial_seed()) % (2**31))
         self.graph: Optional[torch.cuda.CUDAGraph] = None
+        self.capture_error: Optional[str] = None
         self._capture(warmup_iters)
 
     def _batch_from_idx(self):
@@ -67,7 +68,8 @@ class GraphedSACStep:
             with torch.cuda.graph(g):
                 self._eager()
             self.graph = g
-        except Exception:
+        except Exception as e:  # pragma: no cover - GPU only
+            self.capture_error = repr(e)
             self.graph = None  # eager fallback
 
     @property

@@ -21,7 +21,16 @@ def masked_softmax(logits: torch.Tensor, mask: Optional[torch.Tensor]) -> torch.
 
 def sample_categorical(logits: torch.Tensor,
                        mask: Optional[torch.Tensor]) -> Tuple[torch.Tensor, torch.Tensor]:
-    probs = masked_softmax(logits, mask)
-    dist = torch.distributions.Categorical(probs=probs)
-    a = dist.sample()
-    return a, dist.log_prob(a)
+    # Gumbel-max sampling: exact categorical draw with no host-side
+    # validation or synchronization (hipGraph-capturable), equivalent in
+    # distribution to torch.distributions.Categorical(probs).sample()
+    if mask is not None:
+        mask = mask.to(dtype=torch.bool, device=logits.device)
+        neg_inf = torch.finfo(logits.dtype).min
+        logits = torch.where(mask, logits, torch.full_like(logits, neg_inf))
+    logp_all = torch.log_softmax(logits, dim=-1)
+    u = torch.rand_like(logits).clamp_min(1e-20)
+    gumbel = -torch.log(-torch.log(u))
+    a = torch.argmax(logp_all + gumbel, dim=-1)
+    logp = logp_all.gather(-1, a.unsqueeze(-1)).squeeze(-1)
+    return a, logp
