@@ -44,8 +44,11 @@ class MaskedHybridSAC(nn.Module):
                                                 actor.n_g, n_quantiles)
         self.target_critic.load_state_dict(self.critic.state_dict())
         self.nq = n_quantiles
-        self.taus = torch.linspace(1.0 / (2 * n_quantiles),
-                                   1 - 1.0 / (2 * n_quantiles), n_quantiles)
+        # buffer so Module.to() moves it with the model (a plain attribute
+        # would stay on CPU and force a pageable H2D copy per train step —
+        # slow, and illegal inside hipGraph capture)
+        self.register_buffer("taus", torch.linspace(
+            1.0 / (2 * n_quantiles), 1 - 1.0 / (2 * n_quantiles), n_quantiles))
         self.device = device or torch.device("cpu")
         self.log_alpha = torch.tensor(math.log(alpha), requires_grad=True,
                                       device=self.device)
