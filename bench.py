@@ -40,17 +40,22 @@ def parse_args():
     return p.parse_args()
 
 
-def measure_rl_steps_per_sec(device, n_updates=60, batch=256):
-    """CHSAC-AF train-step rate on the paper obs space (49-dim, 8 DC, 8 g)."""
+def measure_rl_steps_per_sec(device, n_updates=200, batch=256):
+    """CHSAC-AF train-step rate on the paper obs space (49-dim, 8 DC, 8 g),
+    using the production path: hipGraph-captured step when available
+    (rl/graphed.py), eager otherwise."""
     import torch
 
     from distributed_cluster_gpus_amd.rl.agent import CHSACAgent, CHSACAgentConfig
+    from distributed_cluster_gpus_amd.rl.graphed import GraphedSACStep
     from distributed_cluster_gpus_amd.rl.replay import ReplayRing
     torch.manual_seed(0)
     obs_dim = 49
+    on_gpu = device.type == "cuda"
     agent = CHSACAgent(CHSACAgentConfig(
         obs_dim=obs_dim, n_dc=8, n_g_choices=8,
-        constraints={"latency_p99": 500.0, "gpu_over": 0.0}, device=str(device)))
+        constraints={"latency_p99": 500.0, "gpu_over": 0.0},
+        device=str(device), graph_capturable=on_gpu))
     ring = ReplayRing(capacity=4096, obs_dim=obs_dim, n_costs=2,
                       cost_names=["latency_p99", "gpu_over"], n_dc=8, n_g=8,
                       device=str(device), seed=0)
@@ -59,14 +64,26 @@ def measure_rl_steps_per_sec(device, n_updates=60, batch=256):
                    a_dc=torch.randint(0, 8, (B,)), a_g=torch.randint(0, 8, (B,)),
                    r=torch.randn(B), costs=torch.rand(B, 2) * 100,
                    done=torch.ones(B))
+    stepper = None
+    if on_gpu:
+        stepper = GraphedSACStep(agent, ring, batch)
+        if not stepper.captured:
+            stepper = None
+
+    def one_step():
+        if stepper is not None:
+            stepper.step()
+        else:
+            agent.train_step(ring.sample(batch), compute_stats=False)
+
     for _ in range(10):  # warmup
-        agent.train_step(ring.sample(batch))
-    if device.type == "cuda":
+        one_step()
+    if on_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(n_updates):
-        agent.train_step(ring.sample(batch))
-    if device.type == "cuda":
+        one_step()
+    if on_gpu:
         torch.cuda.synchronize()
     return n_updates / (time.perf_counter() - t0)
 
