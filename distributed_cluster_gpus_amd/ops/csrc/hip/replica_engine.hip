@@ -215,6 +215,12 @@ __device__ __forceinline__ void store_fence() {
   asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
 }
 
+__device__ __forceinline__ void lds_fence() {
+  // lighter fence for sites that only wrote LDS: waiting lgkmcnt alone
+  // avoids draining unrelated outstanding global loads (vmcnt)
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
 // ---------------- hot per-replica state (LDS-resident) ----------------
 // The fields every event touches live in LDS for the kernel's lifetime
 // (~850 B per replica, ~3.4 KB per 4-replica workgroup of the CU's 160 KiB)
@@ -286,7 +292,7 @@ __device__ void rescan_dc_min(Ctx& c, int d) {
     c.hs->dc_minf[d] = best;
     c.hs->dc_mins[d] = best < D_INF ? slot : -1;
   }
-  store_fence();
+  lds_fence();
 }
 
 // start a job on DC d with (n, f); assumes free >= 1; wave-cooperative.
@@ -352,7 +358,7 @@ __device__ int heuristic_alloc(Ctx& c, int d, int jt) {
     } else nf = fmax(cf, S.dvfs_low);
   }
   if (c.lane == 0) c.hs->cur_freq[d] = (float)nf;
-  store_fence();
+  lds_fence();
   return max(1, g);
 }
 
@@ -512,7 +518,7 @@ __device__ void accrue_to(Ctx& c, double t) {
       c.hs->energy[c.lane] += c.dc_power(c.lane) * dt;
     }
   }
-  store_fence();
+  lds_fence();
 }
 
 // emit one cluster-log row set (logging replica only; wave-cooperative counts)
@@ -923,7 +929,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     }
     if (lane == 0) h->next_log = S.next_log[c.r];
   }
-  store_fence();
+  store_fence();  // global loads (vmcnt) AND the LDS writes (lgkmcnt)
 
   // ---- CHSAC: resume a paused action request ----
   if (ALGO == A_CHSAC && S.pend_kind[c.r] != PEND_NONE) {
@@ -1086,7 +1092,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           }
         }
         if (lane == 0) c.hs->arr_next[idx] = t_min + ia_rl;
-        store_fence();
+        lds_fence();
         rl_request(c, PEND_ARRIVAL, t_min, jt, ing, (float)size, 0.0f, jid,
                    -1, 0);
         paused = true;
@@ -1170,7 +1176,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         }
       }
       if (lane == 0) c.hs->arr_next[idx] = t_min + ia;
-      store_fence();
+      lds_fence();
 
     } else if (kind == 1) {
       // ===== WAN transfer complete: admission =====
@@ -1182,7 +1188,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int jid = S.x_jid[at];
       int ing = S.x_ing[at];
       if (lane == 0) c.l_xt[idx] = D_INF;
-      store_fence();
+      lds_fence();
       if (c.free_gpus(d) > 0) {
         if (ALGO == A_CHSAC && S.x_has_rl[at]) {
           // RL-chosen n (clamped), energy-optimal f (reference :646-667)
@@ -1320,10 +1326,10 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       if (lane < S.n_dc) {
         c.hs->acc_unit[lane] += c.hs->sum_tpt[lane] * S.log_interval;
       }
-      store_fence();
+      lds_fence();
       emit_cluster_rows(c, t_min);
       if (lane == 0) c.hs->next_log = t_min + S.log_interval;
-      store_fence();
+      lds_fence();
     }
   }
 
