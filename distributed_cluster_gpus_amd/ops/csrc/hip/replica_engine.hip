@@ -1101,14 +1101,18 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       // routing
       int d_sel;
       if (ALGO == A_ECO_ROUTE) {
-        // lanes 0..n_dc-1 each score their DC (serial 64-candidate loop)
+        // all 64 lanes score: lane = d*8 + (n-1) covers (DC, n); each lane
+        // reduces over the frequency ladder, then an 8-lane subgroup min per
+        // DC and a wave argmin pick the DC (first-minimum = lowest DC index,
+        // matching the scalar scan order).
         double score = D_INF;
-        if (lane < S.n_dc) {
-          int d = lane;
-          const double* pcf = c.pc3(d, jt);
-          const double* tcf = c.lc3(d, jt);
-          double best = D_INF;
-          for (int n = 1; n <= S.max_gpj; ++n)
+        {
+          int d = lane >> 3;
+          int n = (lane & 7) + 1;
+          if (d < S.n_dc && n <= S.max_gpj) {
+            const double* pcf = c.pc3(d, jt);
+            const double* tcf = c.lc3(d, jt);
+            double best = D_INF;
             for (int q = 0; q < S.n_freq; ++q) {
               double f = S.freq_levels[q];
               double T = d_unit_time(n, f, tcf);
@@ -1119,13 +1123,20 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
               else sc = E;
               if (sc < best) best = sc;
             }
-          if (S.eco_obj == 1) score = (best * size);          // relative order ok
-          else if (S.eco_obj == 2) score = (best * size);
-          else score = best * size;
+            score = best * size;  // relative order preserved per objective
+          }
         }
+        // min within each DC's 8-lane subgroup
+#pragma unroll
+        for (int off = 1; off < 8; off <<= 1)
+          score = fmin(score, __shfl_xor(score, off, 64));
+        // now every lane of a DC group holds its DC's best; keep only the
+        // group leader's value for the cross-DC argmin (tie-break: lowest
+        // lane == lowest DC index)
+        if (lane & 7) score = D_INF;
         int dl;
         wave_argmin_f64(score, dl);
-        d_sel = dl;
+        d_sel = dl >> 3;
       } else {
         d_sel = (int)rbelow(c.rng, (uint32_t)S.n_dc);
       }
