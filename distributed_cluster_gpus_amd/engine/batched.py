@@ -363,15 +363,24 @@ class BatchedEngine:
         launches = 0
         while True:
             self._sim.advance(self.end_time, self.events_per_launch)
-            torch.cuda.synchronize(self.device)
             launches += 1
-            err = int(t["err"].max().item())
+            # one fused D2H status read per launch (err / done / pending /
+            # transitions) instead of several .item() synchronizations
+            status = torch.stack([
+                t["err"].max(),
+                t["done"].min(),
+                (t["req_flag"] == 1).sum().to(torch.int32) if self.is_rl
+                else torch.zeros((), dtype=torch.int32, device=self.device),
+                t["tr_count"][0] if self.is_rl
+                else torch.zeros((), dtype=torch.int32, device=self.device),
+            ]).cpu()
+            err = int(status[0])
             if err != 0:
                 raise RuntimeError(f"batched engine error flags: {err:#x} "
                                    f"(queue/transfer/slot/log overflow)")
             if self.is_rl:
-                self._rl_service()
-            if bool(t["done"].min().item() == 1):
+                self._rl_service(n_req=int(status[2]), n_tr=int(status[3]))
+            if int(status[1]) == 1:
                 break
             if launches > 1000000:
                 raise RuntimeError("batched engine failed to converge")
@@ -387,7 +396,7 @@ class BatchedEngine:
         b = bytes_tensor.to(torch.int64).unsqueeze(1)
         return (b >> torch.arange(width, device=b.device).unsqueeze(0)) & 1 > 0
 
-    def _rl_service(self):
+    def _rl_service(self, n_req=None, n_tr=None):
         """Serve pending action requests with ONE batched policy forward,
         drain completed transitions into the replay ring, and run SAC train
         steps at the configured cadence (the reference trains once per job
@@ -396,7 +405,8 @@ class BatchedEngine:
         cadence is recovered)."""
         t = self.t
         pend = t["req_flag"] == 1
-        n_req = int(pend.sum().item())
+        if n_req is None:
+            n_req = int(pend.sum().item())
         if n_req > 0:
             idx = pend.nonzero(as_tuple=True)[0]
             obs = t["req_obs"][idx]
@@ -413,7 +423,8 @@ class BatchedEngine:
             t["resp_g"][idx] = a["g"].to(torch.int32)
             t["req_flag"][idx] = 2  # REQ_READY
         # transitions -> replay
-        n_tr = int(t["tr_count"].item())
+        if n_tr is None:
+            n_tr = int(t["tr_count"].item())
         if n_tr > 0:
             n_tr = min(n_tr, int(t["tr_s0"].shape[0]))
             costs = t["tr_costs"][:n_tr]
