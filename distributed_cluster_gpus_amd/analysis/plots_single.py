@@ -2,10 +2,8 @@
 plot_single_algo.py main():271-332 — per-DC queues, utilization, busy GPUs,
 energy, the (n, f) operating-point trend, ingress statistics and the
 ingress->DC routing heatmap)."""
-import os
 from typing import List, Optional
 
-import pandas as pd
 
 from .aggregate import load_run
 from .render import emit

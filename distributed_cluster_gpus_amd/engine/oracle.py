@@ -22,7 +22,6 @@ Engine-behaviour notes (quirks preserved deliberately):
 """
 import heapq
 import itertools
-import math
 import os
 import random
 from typing import Dict, List, Optional, Tuple
@@ -30,7 +29,6 @@ from typing import Dict, List, Optional, Tuple
 from ..models.arrivals import ArrivalProcess, sample_job_size
 from ..models.cluster import DataCenterState, JobState, PreemptedJobState
 from ..models.latency import unit_time_s
-from ..models.power import job_power_w
 from ..models.scenario import PAYLOAD_GB, Scenario
 from ..policies.bandit import UCB1DVFS
 from ..policies.gridsearch import best_energy_freq, best_nf_grid, energy_tuple

@@ -99,7 +99,7 @@ class CHSACAgent:
         fused flat gradient all-reduce after each backward, and cross-rank
         averaging of constraint-cost means so the PID-lambda state stays
         identical on every replica."""
-        from ..parallel.dist import allreduce_gradients, allreduce_scalar, world_size
+        from ..parallel.dist import allreduce_gradients, world_size
         self.algo.allreduce_hook = allreduce_gradients
         w = world_size()
         from ..parallel.dist import allreduce_tensor_sum

@@ -16,7 +16,6 @@ Extensions over the reference:
 """
 import argparse
 import os
-import sys
 
 
 def parse_args(argv=None):

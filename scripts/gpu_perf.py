@@ -8,7 +8,6 @@ import time
 import os, sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-import torch
 
 from distributed_cluster_gpus_amd.configs.paper import build_arrivals, paper_scenario
 from distributed_cluster_gpus_amd.engine.batched import BatchedEngine

@@ -5,7 +5,6 @@ import os
 import sys
 import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
-import torch
 from distributed_cluster_gpus_amd.configs.paper import paper_scenario
 from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
 from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess

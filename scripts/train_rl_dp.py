@@ -44,8 +44,7 @@ def main(argv=None):
     from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
     from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
     from distributed_cluster_gpus_amd.parallel.dist import (broadcast_module,
-                                                            init_distributed,
-                                                            world_size)
+                                                            init_distributed)
 
     rank, world = init_distributed()
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
