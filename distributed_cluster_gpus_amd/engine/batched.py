@@ -61,7 +61,7 @@ class BatchedEngine:
                  rl_device: str = 'cuda', rl_batch: int = 256,
                  rl_warmup: int = 1000, rl_buffer: int = 200000,
                  rl_train_interval: int = 256, rl_agent=None,
-                 tr_cap: int = 262144, **_unused):
+                 tr_cap: int = 262144, arrival_trace=None, **_unused):
         if algo not in ALGOS:
             raise ValueError(f"unknown algo {algo!r}")
         if not torch.cuda.is_available():
@@ -188,12 +188,26 @@ class BatchedEngine:
         t["jl_count"] = torch.zeros(1, **i32)
         t["jl_rows"] = torch.zeros((jl_cap, 10), **f64)
 
-        # seed the initial arrival times on host (one inf + one trn per
-        # ingress per replica), Philox-consistent with the device streams:
-        # the kernel's first draws start at ctr = n_streams; host uses
-        # ctr = stream index for the seed draws.
-        arr_np = self._seed_arrivals(arrival_inf, arrival_trn, seed, shard)
-        t["arr_next"].copy_(torch.as_tensor(arr_np, dtype=torch.float64))
+        # arrival-trace replay mode (exact single-replica parity testing):
+        # arrivals come from a recorded (time, size) FIFO per stream
+        self.trace_mode = arrival_trace is not None
+        if self.trace_mode:
+            tt, ts = arrival_trace  # [R][NS][cap] float64 times / float32 sizes
+            tt = np.asarray(tt, np.float64)
+            ts = np.asarray(ts, np.float32)
+            assert tt.shape[0] == R and tt.shape[1] == n_ing * 2
+            self.trace_cap = int(tt.shape[2])
+            t["trace_time"] = torch.as_tensor(tt).to(dev)
+            t["trace_size"] = torch.as_tensor(ts).to(dev)
+            t["trace_pos"] = torch.zeros((R, n_ing * 2), **i32)
+            t["arr_next"].copy_(t["trace_time"][:, :, 0])
+        else:
+            # seed the initial arrival times on host (one inf + one trn per
+            # ingress per replica), Philox-consistent with the device streams:
+            # the kernel's first draws start at ctr = n_streams; host uses
+            # ctr = stream index for the seed draws.
+            arr_np = self._seed_arrivals(arrival_inf, arrival_trn, seed, shard)
+            t["arr_next"].copy_(torch.as_tensor(arr_np, dtype=torch.float64))
         t["rng_ctr"].fill_(n_ing * 2)  # host consumed one block per stream
 
         # ===== CHSAC-AF (RL-in-the-loop) =====
@@ -305,6 +319,8 @@ class BatchedEngine:
             "cl_cap": cl_cap, "jl_cap": jl_cap,
             "obs_dim": obs_dim, "sla_p99_ms": float(sla_p99_ms),
             "tr_cap": int(tr_cap) if self.is_rl else 0,
+            "trace_mode": int(self.trace_mode),
+            "trace_cap": getattr(self, "trace_cap", 0),
         }
         self._sim = self._mod.BatchedSimHip(t, cfg)
         self.meter = ThroughputMeter()

@@ -62,7 +62,8 @@ class OracleEngine:
                  rl_device: str = "cpu", rl_batch: int = 256, rl_warmup: int = 1000,
                  rl_buffer: int = 200000, rl_agent=None,
                  logger=None, show_progress: bool = False,
-                 cluster_writer=None, job_writer=None):
+                 cluster_writer=None, job_writer=None,
+                 arrival_recorder=None):
         if algo not in ALGOS:
             raise ValueError(f"unknown algo {algo!r}")
         self.sc = scenario
@@ -79,6 +80,9 @@ class OracleEngine:
         self.energy_budget_j = energy_budget_j
         self.logger = logger or _NullLogger()
         self.show_progress = show_progress
+        # optional list collecting (t, ingress, jtype, size) per arrival —
+        # used to build exact-replay traces for the GPU engine
+        self.arrival_recorder = arrival_recorder
         self.elastic_scaling = bool(elastic_scaling) and algo == "chsac_af"
 
         # CPython Mersenne stream: seeded exactly like the reference
@@ -281,6 +285,8 @@ class OracleEngine:
     def _on_arrival(self, jtype: str, ing_name: str):
         jid = next(self._jid)
         size = sample_job_size(jtype, self.rng)
+        if self.arrival_recorder is not None:
+            self.arrival_recorder.append((self.now, ing_name, jtype, size))
         job = JobState(jid=jid, ingress=ing_name, jtype=jtype, size=size,
                        arrival_time=self.now)
 

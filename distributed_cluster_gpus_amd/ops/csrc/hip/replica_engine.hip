@@ -149,6 +149,14 @@ struct EngineDesc {
   double* cl_rows;                // [cl_cap][16]
   int* jl_count;                  // [1]
   double* jl_rows;                // [jl_cap][10]
+  // ===== arrival-trace replay mode (single-replica exact-parity testing:
+  // arrivals come from a host-recorded (time, size) FIFO per stream instead
+  // of the Philox draws) =====
+  int trace_mode;                 // 0 off, 1 on
+  int trace_cap;                  // entries per stream
+  const double* trace_time;       // [r][n_streams][cap] absolute times
+  const float* trace_size;        // [r][n_streams][cap]
+  int* trace_pos;                 // [r][n_streams] cursor
   // ===== CHSAC-AF (RL-in-the-loop) state; null unless algo == A_CHSAC =====
   int obs_dim;                    // 1 + 6*n_dc
   double sla_p99_ms;
@@ -1066,13 +1074,29 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int jt = idx & 1;
       int jid = S.jid_ctr[c.r] + 1;
       if (lane == 0) S.jid_ctr[c.r] = jid;
-      double size = jt == 0 ? rpareto_inf(c.rng)
-                            : fmax(0.1, rlognormal(c.rng, log(50000.0), 0.4));
+      double size;
+      if (S.trace_mode) {
+        // replay mode: this arrival's size and the stream's next time come
+        // from the recorded trace
+        int64_t tb = ((int64_t)c.r * S.n_streams + idx) * S.trace_cap;
+        int posn = S.trace_pos[(int64_t)c.r * S.n_streams + idx];
+        size = S.trace_size[tb + posn];
+        int nxt = posn + 1;
+        double t_next = nxt < S.trace_cap ? S.trace_time[tb + nxt] : D_INF;
+        if (lane == 0) {
+          S.trace_pos[(int64_t)c.r * S.n_streams + idx] = nxt;
+          c.hs->arr_next[idx] = t_next;
+        }
+        lds_fence();
+      } else {
+        size = jt == 0 ? rpareto_inf(c.rng)
+                       : fmax(0.1, rlognormal(c.rng, log(50000.0), 0.4));
+      }
       if (ALGO == A_CHSAC) {
         // schedule the next arrival first (RNG order differs from the scalar
         // engines; distributionally identical), then pause for the policy
         double ia_rl = D_INF;
-        {
+        if (!S.trace_mode) {
           double rate = S.arr_rate[jt];
           int mode = S.arr_mode[jt];
           double amp = S.arr_amp[jt];
@@ -1091,7 +1115,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
             }
           }
         }
-        if (lane == 0) c.hs->arr_next[idx] = t_min + ia_rl;
+        if (lane == 0 && !S.trace_mode) c.hs->arr_next[idx] = t_min + ia_rl;
         lds_fence();
         rl_request(c, PEND_ARRIVAL, t_min, jt, ing, (float)size, 0.0f, jid,
                    -1, 0);
@@ -1165,9 +1189,9 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       }
       store_fence();
       // next arrival for this stream (faithful non-accumulating thinning;
-      // reference arrivals.py:35-48)
+      // reference arrivals.py:35-48); in replay mode it was already set
       double ia = D_INF;
-      {
+      if (!S.trace_mode) {
         double rate = S.arr_rate[jt];
         int mode = S.arr_mode[jt];
         double amp = S.arr_amp[jt];
@@ -1186,7 +1210,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           }
         }
       }
-      if (lane == 0) c.hs->arr_next[idx] = t_min + ia;
+      if (lane == 0 && !S.trace_mode) c.hs->arr_next[idx] = t_min + ia;
       lds_fence();
 
     } else if (kind == 1) {
@@ -1464,6 +1488,15 @@ class BatchedSimHip {
     T_PTR(sum_lat, double); T_PTR(sum_lat_inf, double); T_PTR(sum_wait, double);
     T_PTR(cl_count, int); T_PTR(cl_rows, double);
     T_PTR(jl_count, int); T_PTR(jl_rows, double);
+
+    // arrival-trace replay mode
+    S_.trace_mode = cfg.contains("trace_mode") ? cfg["trace_mode"].cast<int>() : 0;
+    S_.trace_cap = cfg.contains("trace_cap") ? cfg["trace_cap"].cast<int>() : 0;
+    if (S_.trace_mode) {
+      T_CPTR(trace_time, double);
+      T_CPTR(trace_size, float);
+      T_PTR(trace_pos, int);
+    }
 
     // CHSAC-AF extras
     S_.obs_dim = cfg.contains("obs_dim") ? cfg["obs_dim"].cast<int>() : 0;

@@ -333,3 +333,75 @@ def test_chsac_batched_via_cli(tmp_path):
                        capture_output=True, text=True, timeout=600, cwd=REPO)
     assert r.returncode == 0, r.stderr[-2000:]
     assert os.path.exists(os.path.join(out, "job_log.csv"))
+
+
+@needs_gpu
+def test_single_replica_exact_trajectory_parity(tmp_path):
+    """SURVEY §4 (c), exact form: one GPU replica fed the ORACLE's recorded
+    arrival stream must reproduce the scalar trajectory event-for-event —
+    identical job log (jid/dc/n/f/times) and matching cluster-log
+    energy/util columns (fp-accumulation tolerance only).  eco_route makes
+    every routing/allocation decision deterministic."""
+    from distributed_cluster_gpus_amd.configs.paper import build_arrivals, paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.engine.oracle import OracleEngine
+
+    duration = 90.0
+    sc = paper_scenario()
+    inf, trn = build_arrivals()
+    rec = []
+    out_o = str(tmp_path / "oracle")
+    OracleEngine(sc, inf, trn, algo="eco_route", duration=duration,
+                 log_interval=5.0, out_dir=out_o, seed=123,
+                 arrival_recorder=rec).run()
+    # build the [1][NS][cap] trace
+    ing_idx = {n: i for i, n in enumerate(sc.ingress_names)}
+    NS = sc.n_ing * 2
+    streams = [[] for _ in range(NS)]
+    for (tt, ing, jtype, size) in rec:
+        s_id = ing_idx[ing] * 2 + (0 if jtype == "inference" else 1)
+        streams[s_id].append((tt, size))
+    cap = max(len(x) for x in streams) + 1
+    times = np.full((1, NS, cap), 1e300)
+    sizes = np.zeros((1, NS, cap), np.float32)
+    for s_id, entries in enumerate(streams):
+        for k, (tt, size) in enumerate(entries):
+            times[0, s_id, k] = tt
+            sizes[0, s_id, k] = size
+
+    sc2 = paper_scenario()
+    out_g = str(tmp_path / "gpu")
+    eng = BatchedEngine(sc2, inf, trn, algo="eco_route", replicas=1,
+                        duration=duration, log_interval=5.0, out_dir=out_g,
+                        seed=999,  # RNG unused in replay+eco mode
+                        enable_logs=True, arrival_trace=(times, sizes))
+    eng.run()
+
+    import pandas as pd
+    jo = pd.read_csv(os.path.join(out_o, "job_log.csv"))
+    jg = pd.read_csv(os.path.join(out_g, "job_log.csv"))
+    assert len(jo) == len(jg), f"job count {len(jo)} vs {len(jg)}"
+    jo = jo.sort_values("jid").reset_index(drop=True)
+    jg = jg.sort_values("jid").reset_index(drop=True)
+    for col in ("jid", "ingress", "type", "dc", "n_gpus"):
+        assert (jo[col] == jg[col]).all(), f"column {col} diverged"
+    for col, tol in (("size", 1e-4), ("f_used", 1e-9), ("net_lat_s", 1e-6),
+                     ("start_s", 1e-6), ("finish_s", 1e-6), ("latency_s", 1e-6),
+                     ("T_pred", 1e-9), ("P_pred", 1e-6), ("E_pred", 1e-6)):
+        d = (jo[col] - jg[col]).abs().max()
+        assert d <= tol, f"column {col} max diff {d}"
+    co = pd.read_csv(os.path.join(out_o, "cluster_log.csv"))
+    cg = pd.read_csv(os.path.join(out_g, "cluster_log.csv"))
+    assert len(co) == len(cg)
+    key = ["time_s", "dc"]
+    m = co.merge(cg, on=key, suffixes=("_o", "_g"))
+    assert len(m) == len(co)
+    for col in ("busy", "free", "run_total", "run_inf", "run_train",
+                "q_inf", "q_train"):
+        assert (m[f"{col}_o"] == m[f"{col}_g"]).all(), f"cluster {col} diverged"
+    for col, rtol in (("power_W", 1e-6), ("energy_kJ", 1e-6),
+                      ("util_inst", 1e-6), ("util_avg", 1e-6),
+                      ("acc_job_unit", 1e-5), ("freq", 1e-9)):
+        a, b = m[f"{col}_o"], m[f"{col}_g"]
+        d = ((a - b).abs() / (a.abs() + 1e-9)).max()
+        assert d <= rtol, f"cluster {col} rel diff {d}"
