@@ -385,10 +385,12 @@ def test_single_replica_exact_trajectory_parity(tmp_path):
     jg = jg.sort_values("jid").reset_index(drop=True)
     for col in ("jid", "ingress", "type", "dc", "n_gpus"):
         assert (jo[col] == jg[col]).all(), f"column {col} diverged"
-    # size is stored f32 on the GPU (CSV prints %.4f): relative tolerance at
-    # f32 epsilon; everything else is f64-exact modulo accumulation order
-    d = ((jo["size"] - jg["size"]).abs() / (jo["size"].abs() + 1.0)).max()
-    assert d <= 2.4e-7, f"size rel diff {d}"
+    # size is stored f32 on the GPU and both sides print %.4f: allow one
+    # print quantum plus f32 epsilon; everything else is f64-exact modulo
+    # accumulation order
+    d = ((jo["size"] - jg["size"]).abs() -
+         (1.01e-4 + 2.4e-7 * jo["size"].abs())).max()
+    assert d <= 0, f"size diff beyond f32+print tolerance by {d}"
     for col, tol in (("f_used", 1e-9), ("net_lat_s", 1e-6),
                      ("start_s", 1e-6), ("finish_s", 1e-6), ("latency_s", 1e-6),
                      ("T_pred", 1e-9), ("P_pred", 1e-6), ("E_pred", 1e-6)):
