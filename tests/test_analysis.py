@@ -88,3 +88,19 @@ def test_plot_cli(two_runs, tmp_path):
     arts2 = plot_single.main(["--run", two_runs["default_policy"],
                               "--out", out2])
     assert len(arts2) == len(SINGLE_FIGURES)
+
+
+def test_plot_sh_wrapper(two_runs, tmp_path):
+    """plot.sh discovers run subdirectories and emits the full report."""
+    import shutil
+    import subprocess
+    import sys
+    root = str(tmp_path / "runs_root")
+    os.makedirs(root)
+    for name, d in two_runs.items():
+        shutil.copytree(d, os.path.join(root, name))
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(["bash", os.path.join(repo, "plot.sh"), root],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert os.path.exists(os.path.join(root, "report", "summary.csv"))
