@@ -192,13 +192,20 @@ class BatchedEngine:
         # arrivals come from a recorded (time, size) FIFO per stream
         self.trace_mode = arrival_trace is not None
         if self.trace_mode:
-            tt, ts = arrival_trace  # [R][NS][cap] float64 times / float32 sizes
+            # (times, sizes[, routed_dcs]) — routed_dcs entries of -1 let the
+            # algorithm's own routing run (eco_route); >=0 replays the
+            # recorded DC choice (exact-parity for random-routing algos)
+            tt, ts = arrival_trace[0], arrival_trace[1]
+            td = arrival_trace[2] if len(arrival_trace) > 2 else \
+                np.full(np.asarray(tt).shape, -1, np.int8)
             tt = np.asarray(tt, np.float64)
             ts = np.asarray(ts, np.float32)
+            td = np.asarray(td, np.int8)
             assert tt.shape[0] == R and tt.shape[1] == n_ing * 2
             self.trace_cap = int(tt.shape[2])
             t["trace_time"] = torch.as_tensor(tt).to(dev)
             t["trace_size"] = torch.as_tensor(ts).to(dev)
+            t["trace_dc"] = torch.as_tensor(td).to(dev)
             t["trace_pos"] = torch.zeros((R, n_ing * 2), **i32)
             t["arr_next"].copy_(t["trace_time"][:, :, 0])
         else:

@@ -285,8 +285,10 @@ class OracleEngine:
     def _on_arrival(self, jtype: str, ing_name: str):
         jid = next(self._jid)
         size = sample_job_size(jtype, self.rng)
+        _rec = None
         if self.arrival_recorder is not None:
-            self.arrival_recorder.append((self.now, ing_name, jtype, size))
+            _rec = [self.now, ing_name, jtype, size, None]
+            self.arrival_recorder.append(_rec)
         job = JobState(jid=jid, ingress=ing_name, jtype=jtype, size=size,
                        arrival_time=self.now)
 
@@ -316,6 +318,8 @@ class OracleEngine:
             dc_name = self.rng.choice(self._dc_names)
             lnet, bw, cost, transfer_s = self._net_tuple(ing_name, dc_name, job)
 
+        if _rec is not None:
+            _rec[4] = dc_name
         self._schedule(self.now + transfer_s, "xfer_done", {
             "ing": ing_name, "dc": dc_name, "jid": jid, "job": job,
             "net_lat_s": lnet, "net_bw_gbps": bw, "net_path_cost_gb": cost})

@@ -156,6 +156,8 @@ struct EngineDesc {
   int trace_cap;                  // entries per stream
   const double* trace_time;       // [r][n_streams][cap] absolute times
   const float* trace_size;        // [r][n_streams][cap]
+  const char* trace_dc;           // [r][n_streams][cap] routed DC (-1 = use
+                                  //   the algorithm's own routing)
   int* trace_pos;                 // [r][n_streams] cursor
   // ===== CHSAC-AF (RL-in-the-loop) state; null unless algo == A_CHSAC =====
   int obs_dim;                    // 1 + 6*n_dc
@@ -1075,12 +1077,14 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int jid = S.jid_ctr[c.r] + 1;
       if (lane == 0) S.jid_ctr[c.r] = jid;
       double size;
+      int trace_d = -1;
       if (S.trace_mode) {
-        // replay mode: this arrival's size and the stream's next time come
-        // from the recorded trace
+        // replay mode: this arrival's size, routed DC and the stream's next
+        // time come from the recorded trace
         int64_t tb = ((int64_t)c.r * S.n_streams + idx) * S.trace_cap;
         int posn = S.trace_pos[(int64_t)c.r * S.n_streams + idx];
         size = S.trace_size[tb + posn];
+        trace_d = S.trace_dc[tb + posn];
         int nxt = posn + 1;
         double t_next = nxt < S.trace_cap ? S.trace_time[tb + nxt] : D_INF;
         if (lane == 0) {
@@ -1124,7 +1128,9 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       }
       // routing
       int d_sel;
-      if (ALGO == A_ECO_ROUTE) {
+      if (S.trace_mode && trace_d >= 0) {
+        d_sel = trace_d;
+      } else if (ALGO == A_ECO_ROUTE) {
         // all 64 lanes score: lane = d*8 + (n-1) covers (DC, n); each lane
         // reduces over the frequency ladder, then an 8-lane subgroup min per
         // DC and a wave argmin pick the DC (first-minimum = lowest DC index,
@@ -1495,6 +1501,7 @@ class BatchedSimHip {
     if (S_.trace_mode) {
       T_CPTR(trace_time, double);
       T_CPTR(trace_size, float);
+      S_.trace_dc = reinterpret_cast<const char*>(t_["trace_dc"].data_ptr<int8_t>());
       T_PTR(trace_pos, int);
     }
 

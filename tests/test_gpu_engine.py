@@ -336,12 +336,21 @@ def test_chsac_batched_via_cli(tmp_path):
 
 
 @needs_gpu
-def test_single_replica_exact_trajectory_parity(tmp_path):
+@pytest.mark.parametrize("algo,kw,use_own_routing", [
+    ("eco_route", {}, True),           # deterministic routing: GPU routes itself
+    ("default_policy", {}, False),     # random routing: replay recorded DCs
+    ("joint_nf", {}, False),
+    ("bandit", {}, False),
+    ("cap_greedy", {"power_cap": 30000.0}, False),
+    ("debug", {"num_fixed_gpus": 2, "fixed_freq": 0.7}, False),
+])
+def test_single_replica_exact_trajectory_parity(tmp_path, algo, kw,
+                                                use_own_routing):
     """SURVEY §4 (c), exact form: one GPU replica fed the ORACLE's recorded
-    arrival stream must reproduce the scalar trajectory event-for-event —
-    identical job log (jid/dc/n/f/times) and matching cluster-log
-    energy/util columns (fp-accumulation tolerance only).  eco_route makes
-    every routing/allocation decision deterministic."""
+    arrival stream (+ routed DCs for random-routing algorithms) must
+    reproduce the scalar trajectory event-for-event — identical job log
+    (jid/dc/n/f/times) and matching cluster-log energy/util columns
+    (fp-accumulation tolerance only)."""
     from distributed_cluster_gpus_amd.configs.paper import build_arrivals, paper_scenario
     from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
     from distributed_cluster_gpus_amd.engine.oracle import OracleEngine
@@ -351,30 +360,35 @@ def test_single_replica_exact_trajectory_parity(tmp_path):
     inf, trn = build_arrivals()
     rec = []
     out_o = str(tmp_path / "oracle")
-    OracleEngine(sc, inf, trn, algo="eco_route", duration=duration,
+    OracleEngine(sc, inf, trn, algo=algo, duration=duration,
                  log_interval=5.0, out_dir=out_o, seed=123,
-                 arrival_recorder=rec).run()
-    # build the [1][NS][cap] trace
+                 arrival_recorder=rec, **kw).run()
+    # build the [1][NS][cap] trace (times, sizes, routed DCs)
     ing_idx = {n: i for i, n in enumerate(sc.ingress_names)}
+    dc_idx = {n: i for i, n in enumerate(sc.dc_names)}
     NS = sc.n_ing * 2
     streams = [[] for _ in range(NS)]
-    for (tt, ing, jtype, size) in rec:
+    for (tt, ing, jtype, size, dc) in rec:
         s_id = ing_idx[ing] * 2 + (0 if jtype == "inference" else 1)
-        streams[s_id].append((tt, size))
+        streams[s_id].append((tt, size, dc))
     cap = max(len(x) for x in streams) + 1
     times = np.full((1, NS, cap), 1e300)
     sizes = np.zeros((1, NS, cap), np.float32)
+    dcs = np.full((1, NS, cap), -1, np.int8)
     for s_id, entries in enumerate(streams):
-        for k, (tt, size) in enumerate(entries):
+        for k, (tt, size, dc) in enumerate(entries):
             times[0, s_id, k] = tt
             sizes[0, s_id, k] = size
+            if not use_own_routing and dc is not None:
+                dcs[0, s_id, k] = dc_idx[dc]
 
     sc2 = paper_scenario()
     out_g = str(tmp_path / "gpu")
-    eng = BatchedEngine(sc2, inf, trn, algo="eco_route", replicas=1,
+    eng = BatchedEngine(sc2, inf, trn, algo=algo, replicas=1,
                         duration=duration, log_interval=5.0, out_dir=out_g,
-                        seed=999,  # RNG unused in replay+eco mode
-                        enable_logs=True, arrival_trace=(times, sizes))
+                        seed=999,  # RNG unused in replay mode
+                        enable_logs=True,
+                        arrival_trace=(times, sizes, dcs), **kw)
     eng.run()
 
     import pandas as pd
