@@ -1282,15 +1282,19 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           S.sum_lat_inf[c.r] += t_min - start;
         }
         if (ALGO == A_BANDIT) {
-          // reward = -E_pred (energy per unit at the used f)
+          // reward = -E_pred (energy per unit at the used f); match the arm
+          // by NEAREST ladder frequency (f_used is stored f32, so exact f64
+          // comparison would never hit)
           double E = d_job_power(n, fused, c.pc3(d, jt)) * T;
           int64_t ab = ((int64_t)c.r * S.n_dc + d) * 2 + jt;
-          for (int k = 0; k < S.n_freq; ++k)
-            if (S.freq_levels[k] == fused) {
-              S.b_n[ab * S.n_freq + k] += 1;
-              S.b_s[ab * S.n_freq + k] += (float)(-E);
-              break;
-            }
+          int bk = 0;
+          double bd = 1e300;
+          for (int k = 0; k < S.n_freq; ++k) {
+            double diff = fabs(S.freq_levels[k] - fused);
+            if (diff < bd) { bd = diff; bk = k; }
+          }
+          S.b_n[ab * S.n_freq + bk] += 1;
+          S.b_s[ab * S.n_freq + bk] += (float)(-E);
         }
       }
       store_fence();
