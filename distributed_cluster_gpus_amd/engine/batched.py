@@ -131,8 +131,8 @@ class BatchedEngine:
         t["err"] = torch.zeros(R, **i32)
         t["arr_next"] = torch.full((R, n_ing * 2), INF, **f64)
         t["busy"] = torch.zeros((R, n_dc), **i32)
-        t["cur_freq"] = torch.empty((R, n_dc), **f32)
-        t["cur_freq"][:] = torch.as_tensor(scenario.default_freq, dtype=torch.float32,
+        t["cur_freq"] = torch.empty((R, n_dc), **f64)
+        t["cur_freq"][:] = torch.as_tensor(scenario.default_freq, dtype=torch.float64,
                                            device=dev)
         t["energy_j"] = torch.zeros((R, n_dc), **f64)
         t["util_time"] = torch.zeros((R, n_dc), **f64)
@@ -145,15 +145,15 @@ class BatchedEngine:
         t["dc_min_slot"] = torch.full((R, n_dc), -1, **i32)
         t["s_finish"] = torch.full((R, total_slots), INF, **f64)
         t["s_start"] = torch.zeros((R, total_slots), **f64)
-        t["s_size"] = torch.zeros((R, total_slots), **f32)
-        t["s_fused"] = torch.zeros((R, total_slots), **f32)
+        t["s_size"] = torch.zeros((R, total_slots), **f64)
+        t["s_fused"] = torch.zeros((R, total_slots), **f64)
         t["s_netlat"] = torch.zeros((R, total_slots), **f32)
         t["s_jid"] = torch.zeros((R, total_slots), **i32)
         t["s_gpus"] = torch.zeros((R, total_slots), **i16)
         t["s_jtype"] = torch.zeros((R, total_slots), **i8)
         t["s_ing"] = torch.zeros((R, total_slots), **i8)
         t["x_time"] = torch.full((R, tcap), INF, **f64)
-        t["x_size"] = torch.zeros((R, tcap), **f32)
+        t["x_size"] = torch.zeros((R, tcap), **f64)
         t["x_netlat"] = torch.zeros((R, tcap), **f32)
         t["x_jid"] = torch.zeros((R, tcap), **i32)
         t["x_dc"] = torch.zeros((R, tcap), **i8)
@@ -161,7 +161,7 @@ class BatchedEngine:
         t["x_ing"] = torch.zeros((R, tcap), **i8)
         t["q_head"] = torch.zeros((R, n_dc, 2), **i32)
         t["q_len"] = torch.zeros((R, n_dc, 2), **i32)
-        t["q_size"] = torch.zeros((R, n_dc, 2, qcap), **f32)
+        t["q_size"] = torch.zeros((R, n_dc, 2, qcap), **f64)
         # queue aux fields (net latency / jid / ingress) are only consumed by
         # the logging replica's job rows -> single-replica allocation
         t["q_netlat"] = torch.zeros((n_dc, 2, qcap), **f32)
@@ -199,7 +199,7 @@ class BatchedEngine:
             td = arrival_trace[2] if len(arrival_trace) > 2 else \
                 np.full(np.asarray(tt).shape, -1, np.int8)
             tt = np.asarray(tt, np.float64)
-            ts = np.asarray(ts, np.float32)
+            ts = np.asarray(ts, np.float64)
             td = np.asarray(td, np.int8)
             assert tt.shape[0] == R and tt.shape[1] == n_ing * 2
             self.trace_cap = int(tt.shape[2])
@@ -234,7 +234,7 @@ class BatchedEngine:
             t["resp_dc"] = torch.zeros(R, **i32)
             t["resp_g"] = torch.zeros(R, **i32)
             t["pend_kind"] = torch.zeros(R, **i32)
-            t["pend_size"] = torch.zeros(R, **f32)
+            t["pend_size"] = torch.zeros(R, **f64)
             t["pend_netlat"] = torch.zeros(R, **f32)
             t["pend_jid"] = torch.zeros(R, **i32)
             t["pend_ing"] = torch.zeros(R, **i32)

@@ -95,7 +95,7 @@ struct EngineDesc {
   double* arr_next;               // [r][n_streams]  (stream = ing*2 + jtype)
   // per (r, dc)
   int* busy;
-  float* cur_freq;
+  double* cur_freq;
   double* energy_j;
   double* util_time;
   double* util_begin;             // [r][dc] first-event stamp (<0 unset)
@@ -108,8 +108,8 @@ struct EngineDesc {
   // job slots [r][total_slots]
   double* s_finish;               // INF = empty
   double* s_start;
-  float* s_size;
-  float* s_fused;
+  double* s_size;                 // f64: service times must be bitwise-equal
+  double* s_fused;                //   to the scalar engines (exact parity)
   float* s_netlat;
   int* s_jid;
   short* s_gpus;                  // 0 = empty
@@ -117,7 +117,7 @@ struct EngineDesc {
   char* s_ing;
   // in-flight transfers [r][tcap]
   double* x_time;                 // INF = empty
-  float* x_size;
+  double* x_size;
   float* x_netlat;
   int* x_jid;
   char* x_dc;
@@ -127,7 +127,7 @@ struct EngineDesc {
   // exist only for the logging replica ([dc][2][qcap], no replica dim)
   int* q_head;
   int* q_len;
-  float* q_size;                  // [r][dc][2][qcap]
+  double* q_size;                 // [r][dc][2][qcap] (f64 for exact parity)
   float* q_netlat;                // [dc][2][qcap] (log replica only)
   int* q_jid;                     // [dc][2][qcap] (log replica only)
   char* q_ing;                    // [dc][2][qcap] (log replica only)
@@ -155,7 +155,7 @@ struct EngineDesc {
   int trace_mode;                 // 0 off, 1 on
   int trace_cap;                  // entries per stream
   const double* trace_time;       // [r][n_streams][cap] absolute times
-  const float* trace_size;        // [r][n_streams][cap]
+  const double* trace_size;       // [r][n_streams][cap]
   const char* trace_dc;           // [r][n_streams][cap] routed DC (-1 = use
                                   //   the algorithm's own routing)
   int* trace_pos;                 // [r][n_streams] cursor
@@ -171,7 +171,7 @@ struct EngineDesc {
   int* resp_g;                    // [r]
   int* pend_kind;                 // [r] PendKind
   // stashed context for the paused event
-  float* pend_size;               // [r]
+  double* pend_size;              // [r]
   float* pend_netlat;             // [r]
   int* pend_jid;                  // [r]
   int* pend_ing;                  // [r]
@@ -249,7 +249,7 @@ struct Hot {
   double acc_unit[MAX_DC];
   double util_begin[MAX_DC];
   double next_log;
-  float cur_freq[MAX_DC];
+  double cur_freq[MAX_DC];
   int dc_mins[MAX_DC];
   int busy[MAX_DC];
   int n_running[MAX_DC];
@@ -306,7 +306,7 @@ __device__ void rescan_dc_min(Ctx& c, int d) {
 }
 
 // start a job on DC d with (n, f); assumes free >= 1; wave-cooperative.
-__device__ void start_job(Ctx& c, int d, int jt, float size, float netlat,
+__device__ void start_job(Ctx& c, int d, int jt, double size, float netlat,
                           int jid, int ing, int n, double f, double now) {
   const EngineDesc& S = *c.S;
   int64_t base = (int64_t)c.r * S.total_slots;
@@ -329,7 +329,7 @@ __device__ void start_job(Ctx& c, int d, int jt, float size, float netlat,
     c.l_fin[cand] = finish;
     S.s_start[base + cand] = now;
     S.s_size[base + cand] = size;
-    S.s_fused[base + cand] = (float)f;
+    S.s_fused[base + cand] = f;
     S.s_netlat[base + cand] = netlat;
     S.s_jid[base + cand] = jid;
     S.s_gpus[base + cand] = (short)n;
@@ -367,7 +367,7 @@ __device__ int heuristic_alloc(Ctx& c, int d, int jt) {
       g = min(free, S.max_gpj);
     } else nf = fmax(cf, S.dvfs_low);
   }
-  if (c.lane == 0) c.hs->cur_freq[d] = (float)nf;
+  if (c.lane == 0) c.hs->cur_freq[d] = nf;
   lds_fence();
   return max(1, g);
 }
@@ -421,7 +421,7 @@ __device__ void decide_nf(Ctx& c, int d, int jt, float size, double now,
 }
 
 // queue ops (wave-uniform; lane-0 writes)
-__device__ bool queue_push(Ctx& c, int d, int jt, float size, float netlat,
+__device__ bool queue_push(Ctx& c, int d, int jt, double size, float netlat,
                            int jid, int ing) {
   const EngineDesc& S = *c.S;
   int ql = d * 2 + jt;
@@ -446,7 +446,7 @@ __device__ bool queue_push(Ctx& c, int d, int jt, float size, float netlat,
   return true;
 }
 
-__device__ bool queue_push_front(Ctx& c, int d, int jt, float size,
+__device__ bool queue_push_front(Ctx& c, int d, int jt, double size,
                                  float netlat, int jid, int ing) {
   const EngineDesc& S = *c.S;
   int ql = d * 2 + jt;
@@ -472,7 +472,7 @@ __device__ bool queue_push_front(Ctx& c, int d, int jt, float size,
   return true;
 }
 
-__device__ bool queue_pop(Ctx& c, int d, int jt, float& size, float& netlat,
+__device__ bool queue_pop(Ctx& c, int d, int jt, double& size, float& netlat,
                           int& jid, int& ing) {
   const EngineDesc& S = *c.S;
   int ql = d * 2 + jt;
@@ -502,7 +502,8 @@ template <int ALGO>
 __device__ void drain_queues(Ctx& c, int d, double now) {
   const EngineDesc& S = *c.S;
   while (c.free_gpus(d) > 0) {
-    float size, netlat;
+    double size;
+    float netlat;
     int jid, ing;
     int jt;
     if (S.inf_priority && queue_pop(c, d, 0, size, netlat, jid, ing)) jt = 0;
@@ -576,7 +577,7 @@ __device__ void emit_cluster_rows(Ctx& c, double now) {
 }
 
 __device__ void emit_job_row(Ctx& c, int d, int jt, int jid, int ing,
-                             float size, double fused, int n, float netlat,
+                             double size, double fused, int n, float netlat,
                              double start, double finish) {
   const EngineDesc& S = *c.S;
   if (c.r != S.log_replica) return;
@@ -657,7 +658,7 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
       c.hs->p_active[d] += d_job_power(n, best_fto, c.pc3(d, jt)) -
                            d_job_power(n, old_f, c.pc3(d, jt));
       c.hs->sum_tpt[d] += 1.0 / T_new - 1.0 / T_old;
-      S.s_fused[base + best_slot] = (float)best_fto;
+      S.s_fused[base + best_slot] = best_fto;
       c.l_fin[best_slot] = finish_new;
     }
     store_fence();
@@ -744,7 +745,7 @@ __device__ void rl_build_masks(Ctx& c, int& mdc, int& mg) {
 
 // write an action request and stash the paused-event context
 __device__ void rl_request(Ctx& c, int kind, double now, int jt, int ing,
-                           float size, float netlat, int jid, int src_dc,
+                           double size, float netlat, int jid, int src_dc,
                            int from_inf) {
   const EngineDesc& S = *c.S;
   rl_build_obs(c, now, &S.req_obs[(int64_t)c.r * S.obs_dim]);
@@ -785,7 +786,7 @@ __device__ double rl_energy_freq(Ctx& c, int d, int jt, int n) {
 
 // start a job carrying an RL trace; returns chosen slot via start_job's path.
 // (duplicates start_job, then fills the rl trace of the slot just used)
-__device__ void rl_start_job(Ctx& c, int d, int jt, float size, float netlat,
+__device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
                              int jid, int ing, int n, double f, double now,
                              const float* s0, int a_dc, int a_g,
                              int mdc, int mg, int n_rew) {
@@ -812,7 +813,7 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, float size, float netlat,
     c.l_fin[cand] = finish;
     S.s_start[base + cand] = now;
     S.s_size[base + cand] = size;
-    S.s_fused[base + cand] = (float)f;
+    S.s_fused[base + cand] = f;
     S.s_netlat[base + cand] = netlat;
     S.s_jid[base + cand] = jid;
     S.s_gpus[base + cand] = (short)n;
@@ -950,7 +951,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     int pk = S.pend_kind[c.r];
     int jt = S.pend_jt[c.r];
     int ing = S.pend_ing[c.r];
-    float size = S.pend_size[c.r];
+    double size = S.pend_size[c.r];
     float netlat = S.pend_netlat[c.r];
     int jid = S.pend_jid[c.r];
     if (pk == PEND_ARRIVAL) {
@@ -1121,7 +1122,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         }
         if (lane == 0 && !S.trace_mode) c.hs->arr_next[idx] = t_min + ia_rl;
         lds_fence();
-        rl_request(c, PEND_ARRIVAL, t_min, jt, ing, (float)size, 0.0f, jid,
+        rl_request(c, PEND_ARRIVAL, t_min, jt, ing, size, 0.0f, jid,
                    -1, 0);
         paused = true;
         break;
@@ -1186,7 +1187,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       } else if (lane == 0) {
         int64_t at = (int64_t)c.r * S.tcap + cand;
         c.l_xt[cand] = t_min + lnet + xfer;
-        S.x_size[at] = (float)size;
+        S.x_size[at] = size;
         S.x_netlat[at] = (float)lnet;
         S.x_jid[at] = jid;
         S.x_dc[at] = (char)d_sel;
@@ -1224,7 +1225,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int64_t at = (int64_t)c.r * S.tcap + idx;
       int d = S.x_dc[at];
       int jt = S.x_jtype[at];
-      float size = S.x_size[at];
+      double size = S.x_size[at];
       float netlat = S.x_netlat[at];
       int jid = S.x_jid[at];
       int ing = S.x_ing[at];
@@ -1257,7 +1258,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int64_t at = sbase + slot;
       int jt = S.s_jtype[at];
       int n = S.s_gpus[at];
-      float size = S.s_size[at];
+      double size = S.s_size[at];
       double fused = S.s_fused[at];
       double start = S.s_start[at];
       float netlat = S.s_netlat[at];
@@ -1326,7 +1327,8 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         }
         // chsac drains AT MOST ONE queued job per finish, via a fresh policy
         // action (reference :849-890)
-        float qsize, qnetlat;
+        double qsize;
+        float qnetlat;
         int qjid, qing;
         bool popped = false;
         int from_inf = 0;
@@ -1360,7 +1362,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
             if (c.hs->busy[lane] == 0) {
               double fm = S.freq_levels[0];
               for (int k = 1; k < S.n_freq; ++k) fm = fmin(fm, S.freq_levels[k]);
-              c.hs->cur_freq[lane] = (float)fm;
+              c.hs->cur_freq[lane] = fm;
             }
           }
           store_fence();
@@ -1472,21 +1474,21 @@ class BatchedSimHip {
     S_.rng_ctr = reinterpret_cast<uint64_t*>(t_["rng_ctr"].data_ptr<int64_t>());
     T_PTR(jid_ctr, int); T_PTR(done, int); T_PTR(err, int);
     T_PTR(arr_next, double);
-    T_PTR(busy, int); T_PTR(cur_freq, float); T_PTR(energy_j, double);
+    T_PTR(busy, int); T_PTR(cur_freq, double); T_PTR(energy_j, double);
     T_PTR(util_time, double); T_PTR(util_begin, double); T_PTR(acc_unit, double);
     T_PTR(p_active, double); T_PTR(sum_tpt, double); T_PTR(n_running, int);
     T_PTR(dc_min_finish, double); T_PTR(dc_min_slot, int);
-    T_PTR(s_finish, double); T_PTR(s_start, double); T_PTR(s_size, float);
-    T_PTR(s_fused, float); T_PTR(s_netlat, float); T_PTR(s_jid, int);
+    T_PTR(s_finish, double); T_PTR(s_start, double); T_PTR(s_size, double);
+    T_PTR(s_fused, double); T_PTR(s_netlat, float); T_PTR(s_jid, int);
     S_.s_gpus = reinterpret_cast<short*>(t_["s_gpus"].data_ptr<int16_t>());
     S_.s_jtype = reinterpret_cast<char*>(t_["s_jtype"].data_ptr<int8_t>());
     S_.s_ing = reinterpret_cast<char*>(t_["s_ing"].data_ptr<int8_t>());
-    T_PTR(x_time, double); T_PTR(x_size, float); T_PTR(x_netlat, float);
+    T_PTR(x_time, double); T_PTR(x_size, double); T_PTR(x_netlat, float);
     T_PTR(x_jid, int);
     S_.x_dc = reinterpret_cast<char*>(t_["x_dc"].data_ptr<int8_t>());
     S_.x_jtype = reinterpret_cast<char*>(t_["x_jtype"].data_ptr<int8_t>());
     S_.x_ing = reinterpret_cast<char*>(t_["x_ing"].data_ptr<int8_t>());
-    T_PTR(q_head, int); T_PTR(q_len, int); T_PTR(q_size, float);
+    T_PTR(q_head, int); T_PTR(q_len, int); T_PTR(q_size, double);
     T_PTR(q_netlat, float); T_PTR(q_jid, int);
     S_.q_ing = reinterpret_cast<char*>(t_["q_ing"].data_ptr<int8_t>());
     T_PTR(b_n, int); T_PTR(b_s, float);
@@ -1504,7 +1506,7 @@ class BatchedSimHip {
     S_.trace_cap = cfg.contains("trace_cap") ? cfg["trace_cap"].cast<int>() : 0;
     if (S_.trace_mode) {
       T_CPTR(trace_time, double);
-      T_CPTR(trace_size, float);
+      T_CPTR(trace_size, double);
       S_.trace_dc = reinterpret_cast<const char*>(t_["trace_dc"].data_ptr<int8_t>());
       T_PTR(trace_pos, int);
     }
@@ -1516,7 +1518,7 @@ class BatchedSimHip {
     if (S_.algo == A_CHSAC) {
       T_PTR(req_flag, int); T_PTR(req_obs, float); T_PTR(req_mdc, int);
       T_PTR(req_mg, int); T_PTR(resp_dc, int); T_PTR(resp_g, int);
-      T_PTR(pend_kind, int); T_PTR(pend_size, float); T_PTR(pend_netlat, float);
+      T_PTR(pend_kind, int); T_PTR(pend_size, double); T_PTR(pend_netlat, float);
       T_PTR(pend_jid, int); T_PTR(pend_ing, int); T_PTR(pend_jt, int);
       T_PTR(pend_dc, int); T_PTR(pend_from_inf, int);
       T_PTR(slot_s0, float);
