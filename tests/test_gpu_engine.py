@@ -377,7 +377,7 @@ def test_single_replica_exact_trajectory_parity(tmp_path, algo, kw,
         streams[s_id].append((tt, size, dc))
     cap = max(len(x) for x in streams) + 1
     times = np.full((1, NS, cap), 1e300)
-    sizes = np.zeros((1, NS, cap), np.float32)
+    sizes = np.zeros((1, NS, cap), np.float64)
     dcs = np.full((1, NS, cap), -1, np.int8)
     for s_id, entries in enumerate(streams):
         for k, (tt, size, dc) in enumerate(entries):
