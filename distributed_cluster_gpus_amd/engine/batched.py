@@ -169,7 +169,7 @@ class BatchedEngine:
         t["q_ing"] = torch.zeros((n_dc, 2, qcap), **i8)
         nb = 1 if algo != "bandit" else R
         t["b_n"] = torch.zeros((nb, n_dc, 2, n_freq), **i32)
-        t["b_s"] = torch.zeros((nb, n_dc, 2, n_freq), **f32)
+        t["b_s"] = torch.zeros((nb, n_dc, 2, n_freq), **f64)
         t["b_t"] = torch.zeros(R, **i64)
         t["ev_count"] = torch.zeros(R, **i64)
         t["jobs_done"] = torch.zeros(R, **i64)

@@ -133,7 +133,7 @@ struct EngineDesc {
   char* q_ing;                    // [dc][2][qcap] (log replica only)
   // bandit state [r][dc][2][n_freq]
   int* b_n;
-  float* b_s;
+  double* b_s;
   long long* b_t;                 // [r]
   // metrics [r]
   long long* ev_count;
@@ -510,7 +510,22 @@ __device__ void drain_queues(Ctx& c, int d, double now) {
     else if (queue_pop(c, d, 1, size, netlat, jid, ing)) jt = 1;
     else break;
     int n; double f;
-    decide_nf<ALGO>(c, d, jt, size, now, n, f);
+    if (ALGO == A_CARBON_COST) {
+      // the reference's drain path always uses the CARBON objective for
+      // carbon_cost (simulator_paper_multi.py:909-920), unlike its admission
+      // path which prefers cost when a price is set (:622-645)
+      GridPick g = wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                                    S.n_freq, S.max_gpj, 1, S.carbon[d], 0.0,
+                                    false, 0.0);
+      n = g.n; f = g.f;
+    } else if (ALGO == A_JOINT_NF || ALGO == A_BANDIT) {
+      // explicit drain branches in the reference (:892-907); everything else
+      // drains through the heuristic allocator (:922-927) — including debug
+      decide_nf<ALGO>(c, d, jt, size, now, n, f);
+    } else {
+      n = heuristic_alloc(c, d, jt);
+      f = c.hs->cur_freq[d];
+    }
     n = max(1, min(n, c.free_gpus(d)));
     start_job(c, d, jt, size, netlat, jid, ing, n, f, now);
   }
@@ -1295,7 +1310,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
             if (diff < bd) { bd = diff; bk = k; }
           }
           S.b_n[ab * S.n_freq + bk] += 1;
-          S.b_s[ab * S.n_freq + bk] += (float)(-E);
+          S.b_s[ab * S.n_freq + bk] += -E;
         }
       }
       store_fence();
@@ -1491,7 +1506,7 @@ class BatchedSimHip {
     T_PTR(q_head, int); T_PTR(q_len, int); T_PTR(q_size, double);
     T_PTR(q_netlat, float); T_PTR(q_jid, int);
     S_.q_ing = reinterpret_cast<char*>(t_["q_ing"].data_ptr<int8_t>());
-    T_PTR(b_n, int); T_PTR(b_s, float);
+    T_PTR(b_n, int); T_PTR(b_s, double);
     S_.b_t = reinterpret_cast<long long*>(t_["b_t"].data_ptr<int64_t>());
     S_.ev_count = reinterpret_cast<long long*>(t_["ev_count"].data_ptr<int64_t>());
     S_.jobs_done = reinterpret_cast<long long*>(t_["jobs_done"].data_ptr<int64_t>());
