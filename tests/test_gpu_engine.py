@@ -140,16 +140,25 @@ def test_joint_nf_decisions_exact(tmp_path):
 
 @needs_gpu
 def test_debug_pins_nf(tmp_path):
+    """Directly-admitted jobs use the pinned (n, f); queued jobs drain
+    through the heuristic (reference behaviour, :922-927) — use a light load
+    so nothing queues."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="poisson", rate=0.5)
+    trn = ArrivalProcess(mode="off", rate=0.0)
     out = str(tmp_path / "dbg")
-    eng = make_engine(algo="debug", replicas=4, duration=60.0, out_dir=out,
-                      enable_logs=True, num_fixed_gpus=2, fixed_freq=0.7)
+    eng = BatchedEngine(sc, inf, trn, algo="debug", replicas=4, duration=90.0,
+                        log_interval=5.0, out_dir=out, enable_logs=True,
+                        seed=123, num_fixed_gpus=2, fixed_freq=0.7)
     eng.run()
     with open(os.path.join(out, "job_log.csv")) as fh:
         rows = list(csv.DictReader(fh))
     assert rows
-    started_direct = [r for r in rows if r["n_gpus"] == "2"]
-    assert len(started_direct) > 0.9 * len(rows)
-    for r in started_direct:
+    for r in rows:
+        assert r["n_gpus"] == "2"
         assert r["f_used"] == "0.700"
 
 
