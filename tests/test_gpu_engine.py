@@ -430,11 +430,15 @@ def test_single_replica_exact_trajectory_parity(tmp_path, algo, kw,
                 "q_inf", "q_train"):
         assert (m[f"{col}_o"] == m[f"{col}_g"]).all(), f"cluster {col} diverged"
     # power/energy use an incremental active-power cache on the GPU vs the
-    # oracle's per-event recomputation: identical values modulo f64
-    # accumulation order
-    for col, rtol in (("power_W", 1e-8), ("energy_kJ", 1e-8),
-                      ("util_inst", 1e-12), ("util_avg", 1e-10),
-                      ("acc_job_unit", 1e-6), ("freq", 1e-12)):
+    # oracle's per-event recomputation: identical modulo f64 accumulation
+    # order, which can straddle a CSV print-rounding boundary — allow one
+    # print quantum plus the fp-accumulation relative tolerance
+    for col, quantum, rtol in (("power_W", 0.01, 1e-8),
+                               ("energy_kJ", 1e-4, 1e-8),
+                               ("util_inst", 1e-4, 1e-12),
+                               ("util_avg", 1e-4, 1e-10),
+                               ("acc_job_unit", 1e-4, 1e-6),
+                               ("freq", 0.0, 1e-12)):
         a, b = m[f"{col}_o"], m[f"{col}_g"]
-        d = ((a - b).abs() / (a.abs() + 1e-9)).max()
-        assert d <= rtol, f"cluster {col} rel diff {d}"
+        d = ((a - b).abs() - (1.01 * quantum + rtol * a.abs())).max()
+        assert d <= 0, f"cluster {col} beyond print+fp tolerance by {d}"
