@@ -34,7 +34,7 @@ def emit(df: pd.DataFrame, out_dir: str, name: str,
     os.makedirs(out_dir, exist_ok=True)
     csv_path = os.path.join(out_dir, f"{name}.csv")
     df.to_csv(csv_path, index=False)
-    if not have_mpl():
+    if not have_mpl() or len(df) == 0:
         return csv_path
     plt = _plt()
     fig, ax = plt.subplots(figsize=(9, 5))
@@ -70,10 +70,12 @@ def emit(df: pd.DataFrame, out_dir: str, name: str,
                 ax.scatter(df[x], df[y], s=12)
             ax.set_xlabel(x)
         elif kind == "box":
-            groups = [(str(k), sub[y].values) for k, sub in df.groupby(hue)]
-            ax.boxplot([g[1] for g in groups], labels=[g[0] for g in groups],
-                       showfliers=False)
-            ax.tick_params(axis="x", rotation=30)
+            groups = [(str(k), sub[y].values) for k, sub in df.groupby(hue)
+                      if len(sub)]
+            if groups:
+                ax.boxplot([g[1] for g in groups],
+                           labels=[g[0] for g in groups], showfliers=False)
+                ax.tick_params(axis="x", rotation=30)
         elif kind == "heatmap":
             mat = df.set_index(df.columns[0])
             im = ax.imshow(mat.values, aspect="auto", cmap="viridis")
@@ -89,6 +91,8 @@ def emit(df: pd.DataFrame, out_dir: str, name: str,
             ax.set_ylabel(ylabel)
         fig.tight_layout()
         fig.savefig(os.path.join(out_dir, f"{name}.png"), dpi=110)
+    except Exception as e:  # rendering must never kill a report: the CSV
+        print(f"[render] {name}.png skipped: {e}")   # data is the artifact
     finally:
         plt.close(fig)
     return csv_path
