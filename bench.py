@@ -35,6 +35,7 @@ def parse_args():
     p.add_argument("--events-per-step", type=int, default=4000)
     p.add_argument("--algo", type=str, default="default_policy")
     p.add_argument("--seed", type=int, default=123)
+    p.add_argument("--subwave", type=int, default=64, choices=[8, 64])
     p.add_argument("--with-rl", type=int, default=1,
                    help="also measure CHSAC-AF SAC updates/sec (untimed region)")
     return p.parse_args()
@@ -123,7 +124,7 @@ def main():
                         replicas=total_replicas, duration=duration,
                         log_interval=5.0, out_dir=None, seed=args.seed,
                         device=device, rank=rank, world=world, qcap=qcap,
-                        enable_logs=False)
+                        enable_logs=False, subwave=args.subwave)
 
     ev = eng.t["ev_count"]
 

@@ -61,13 +61,16 @@ class BatchedEngine:
                  rl_device: str = 'cuda', rl_batch: int = 256,
                  rl_warmup: int = 1000, rl_buffer: int = 200000,
                  rl_train_interval: int = 256, rl_agent=None,
-                 tr_cap: int = 262144, arrival_trace=None, **_unused):
+                 tr_cap: int = 262144, arrival_trace=None,
+                 subwave: int = 64, **_unused):
         if algo not in ALGOS:
             raise ValueError(f"unknown algo {algo!r}")
         if not torch.cuda.is_available():
             raise RuntimeError("BatchedEngine requires a ROCm GPU "
                                "(no silent CPU fallback)")
-        self._mod = load_sim_hip()  # raises if the gfx950 extension is missing
+        # subwave=64: one wavefront per replica (default);
+        # subwave=8: eight replicas per wavefront (mw variant)
+        self._mod = load_sim_hip("_sim_hip" if subwave == 64 else "_sim_hip_mw")
         self.sc = scenario
         self.algo = algo
         self.device = device or torch.device("cuda", torch.cuda.current_device())

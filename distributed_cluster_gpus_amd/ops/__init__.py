@@ -27,24 +27,25 @@ def have_des_core() -> bool:
         return False
 
 
-def load_sim_hip():
-    """Import the gfx950 batched-engine extension.  torch must be imported
-    first (the .so links against libtorch)."""
+def load_sim_hip(name: str = "_sim_hip"):
+    """Import a gfx950 batched-engine extension ("_sim_hip" = wave-per-replica,
+    "_sim_hip_mw" = 8-replicas-per-wave).  torch must be imported first (the
+    .so links against libtorch)."""
     import importlib.util
     import torch  # noqa: F401  (symbol provider)
-    path = os.path.join(os.path.dirname(__file__), "_sim_hip.so")
+    path = os.path.join(os.path.dirname(__file__), f"{name}.so")
     if not os.path.exists(path):
         raise ImportError(
-            "_sim_hip.so (gfx950 HIP engine) not built. Run "
+            f"{name}.so (gfx950 HIP engine) not built. Run "
             "`python -m distributed_cluster_gpus_amd.ops.build_hip` "
             "(requires hipcc; cross-compiles fine without a GPU).")
     import sys
-    if "_sim_hip" in sys.modules:
-        return sys.modules["_sim_hip"]
-    spec = importlib.util.spec_from_file_location("_sim_hip", path)
+    if name in sys.modules:
+        return sys.modules[name]
+    spec = importlib.util.spec_from_file_location(name, path)
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
-    sys.modules["_sim_hip"] = mod
+    sys.modules[name] = mod
     return mod
 
 

@@ -25,7 +25,7 @@ def torch_paths():
     return inc, lib
 
 
-def build_extension(name: str, sources, verbose=True) -> str:
+def build_extension(name: str, sources, verbose=True, extra_defs=()) -> str:
     inc, lib = torch_paths()
     import sysconfig
     py_inc = sysconfig.get_paths()["include"]
@@ -33,7 +33,7 @@ def build_extension(name: str, sources, verbose=True) -> str:
     cmd = ["hipcc", f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
            "-shared", f"-DTORCH_EXTENSION_NAME={name}", "-DUSE_ROCM",
            "-DGLOG_USE_GLOG_EXPORT",
-           "-Wno-unused-result"]
+           "-Wno-unused-result"] + list(extra_defs)
     for i in inc:
         cmd.append(f"-I{i}")
     cmd.append(f"-I{py_inc}")
@@ -50,7 +50,12 @@ def build_extension(name: str, sources, verbose=True) -> str:
 
 
 def build_all(verbose=True):
-    return [build_extension("_sim_hip", ["replica_engine.hip"], verbose=verbose)]
+    return [
+        build_extension("_sim_hip", ["replica_engine.hip"], verbose=verbose),
+        # multi-replica-per-wave variant: 8 replicas x 8 lanes per wavefront
+        build_extension("_sim_hip_mw", ["replica_engine.hip"], verbose=verbose,
+                        extra_defs=["-DDCG_SUBWAVE=8"]),
+    ]
 
 
 if __name__ == "__main__":

@@ -215,7 +215,7 @@ struct EngineDesc {
 // ---------------- wave helpers ----------------
 __device__ __forceinline__ int wave_sum_i32(int v) {
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  for (int off = SUBW / 2; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
   return v;
 }
 
@@ -238,7 +238,8 @@ __device__ __forceinline__ void lds_fence() {
 // dependent global-memory round trips per event (PMC before: waves waiting
 // 63% of cycles on memory; see profiles/README.md).
 constexpr int MAX_STREAMS = 16;
-constexpr int WAVES_PER_BLOCK = 2;
+constexpr int THREADS_PER_BLOCK = 128;
+constexpr int REPLICAS_PER_BLOCK = THREADS_PER_BLOCK / DCG_SUBWAVE;
 struct Hot {
   double arr_next[MAX_STREAMS];
   double dc_minf[MAX_DC];
@@ -291,7 +292,7 @@ __device__ void rescan_dc_min(Ctx& c, int d) {
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
   double v = D_INF;
   int slot = -1;
-  for (int k = lo + c.lane; k < hi; k += 64) {
+  for (int k = lo + c.lane; k < hi; k += SUBW) {
     double f = c.l_fin[k];
     if (f < v) { v = f; slot = k; }
   }
@@ -313,11 +314,11 @@ __device__ void start_job(Ctx& c, int d, int jt, double size, float netlat,
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
   // find first empty slot (finish == INF <=> s_gpus == 0), from the LDS mirror
   int cand = INT_MAX;
-  for (int k = lo + c.lane; k < hi; k += 64) {
+  for (int k = lo + c.lane; k < hi; k += SUBW) {
     if (c.l_fin[k] >= D_INF) { cand = k; break; }
   }
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1)
+  for (int off = SUBW / 2; off > 0; off >>= 1)
     cand = min(cand, __shfl_xor(cand, off, 64));
   if (cand == INT_MAX) {  // cannot happen if capacities == total_gpus
     if (c.lane == 0) atomicOr(&S.err[c.r], ERR_SLOT_OVF);
@@ -555,7 +556,7 @@ __device__ void emit_cluster_rows(Ctx& c, double now) {
   for (int d = 0; d < S.n_dc; ++d) {
     int lo = S.slot_off[d], hi = S.slot_off[d + 1];
     int cnt_inf = 0;
-    for (int k = lo + c.lane; k < hi; k += 64)
+    for (int k = lo + c.lane; k < hi; k += SUBW)
       if (S.s_gpus[base + k] != 0 && S.s_jtype[base + k] == 0) cnt_inf++;
     cnt_inf = wave_sum_i32(cnt_inf);
     int run_total = c.hs->n_running[d];
@@ -629,7 +630,7 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
     double best_rho = D_INF;
     int best_slot = -1;
     double best_fto = 0;
-    for (int k = c.lane; k < S.total_slots; k += 64) {
+    for (int k = c.lane; k < S.total_slots; k += SUBW) {
       if (c.l_fin[k] >= D_INF) continue;
       int d = S.slot_dc[k];
       double fu = S.s_fused[base + k];
@@ -809,11 +810,11 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
   int64_t base = (int64_t)c.r * S.total_slots;
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
   int cand = INT_MAX;
-  for (int k = lo + c.lane; k < hi; k += 64) {
+  for (int k = lo + c.lane; k < hi; k += SUBW) {
     if (c.l_fin[k] >= D_INF) { cand = k; break; }
   }
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1)
+  for (int off = SUBW / 2; off > 0; off >>= 1)
     cand = min(cand, __shfl_xor(cand, off, 64));
   if (cand == INT_MAX) {
     if (c.lane == 0) atomicOr(&S.err[c.r], ERR_SLOT_OVF);
@@ -822,7 +823,7 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
   double T = d_unit_time(n, f, c.lc3(d, jt));
   double finish = now + (double)size * T;
   // copy s0 trace (lane-parallel over obs_dim)
-  for (int k = c.lane; k < S.obs_dim; k += 64)
+  for (int k = c.lane; k < S.obs_dim; k += SUBW)
     S.slot_s0[(base + cand) * S.obs_dim + k] = s0[k];
   if (c.lane == 0) {
     c.l_fin[cand] = finish;
@@ -864,12 +865,13 @@ __device__ void rl_emit_transition(Ctx& c, const float* s0, int a_dc, int a_g,
   rl_build_obs(c, now, s1);
   int idx = 0;
   if (c.lane == 0) idx = atomicAdd(S.tr_count, 1);
-  idx = __shfl(idx, 0, 64);
+  // broadcast from THIS subgroup's base lane (absolute in-wave index)
+  idx = __shfl(idx, (threadIdx.x & 63) & ~(SUBW - 1), 64);
   if (idx >= S.tr_cap) {
     if (c.lane == 0) atomicOr(&S.err[c.r], ERR_LOG_OVF);
     return;
   }
-  for (int k = c.lane; k < S.obs_dim; k += 64) {
+  for (int k = c.lane; k < S.obs_dim; k += SUBW) {
     S.tr_s0[(int64_t)idx * S.obs_dim + k] = s0[k];
     S.tr_s1[(int64_t)idx * S.obs_dim + k] = s1[k];
   }
@@ -888,15 +890,17 @@ __device__ void rl_emit_transition(Ctx& c, const float* s0, int a_dc, int a_g,
 
 // ---------------- the advance kernel ----------------
 template <int ALGO>
-__global__ void __launch_bounds__(64 * WAVES_PER_BLOCK)
+__global__ void __launch_bounds__(THREADS_PER_BLOCK)
 advance_kernel(EngineDesc S, double t_target, long long max_ev) {
-  int wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  int lane = threadIdx.x & 63;
-  if (wave >= S.n_rep) return;
+  // SUBW lanes form one replica slot (64: wave-per-replica; 8: eight
+  // replicas per wavefront)
+  int slot_id = (blockIdx.x * blockDim.x + threadIdx.x) / SUBW;
+  int lane = threadIdx.x & (SUBW - 1);
+  if (slot_id >= S.n_rep) return;
 
   Ctx c;
   c.S = &S;
-  c.r = wave;
+  c.r = slot_id;
   c.lane = lane;
   if (S.done[c.r]) return;
   // CHSAC fast-path: a replica still waiting for its policy response must
@@ -913,28 +917,38 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   long long n_events = 0;
   bool paused = false;
 
-  // ---- carve the dynamic-LDS region: per wave, Hot + s_finish + x_time ----
-  // (single extern __shared__ object, every carve 16-byte aligned)
+  // ---- carve the dynamic-LDS region ----
+  // SUBW==64: per replica, Hot + s_finish mirror + x_time mirror.
+  // SUBW==8: Hot only (the mirrors would exceed the LDS budget at 8
+  // replicas/wave); l_fin/l_xt then alias the replica's global rows.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   {
     size_t hot_sz = (sizeof(Hot) + 15) & ~size_t(15);
-    size_t fin_sz = (size_t)S.total_slots * sizeof(double);
-    size_t xt_sz = (size_t)S.tcap * sizeof(double);
+    size_t fin_sz = SUBW == 64 ? (size_t)S.total_slots * sizeof(double) : 0;
+    size_t xt_sz = SUBW == 64 ? (size_t)S.tcap * sizeof(double) : 0;
     size_t stride = hot_sz + fin_sz + xt_sz;
-    char* base = smem + (threadIdx.x >> 6) * stride;
+    char* base = smem + (threadIdx.x / SUBW) * stride;
     c.hs = reinterpret_cast<Hot*>(base);
-    c.l_fin = reinterpret_cast<double*>(base + hot_sz);
-    c.l_xt = reinterpret_cast<double*>(base + hot_sz + fin_sz);
+    if (SUBW == 64) {
+      c.l_fin = reinterpret_cast<double*>(base + hot_sz);
+      c.l_xt = reinterpret_cast<double*>(base + hot_sz + fin_sz);
+    } else {
+      c.l_fin = S.s_finish + sbase;
+      c.l_xt = S.x_time + (int64_t)c.r * S.tcap;
+    }
   }
   {
     Hot* h = c.hs;
     int nd = S.n_dc;
-    // slot finish times + transfer times into LDS (lane-strided)
-    for (int k = lane; k < S.total_slots; k += 64)
-      c.l_fin[k] = S.s_finish[sbase + k];
-    for (int k = lane; k < S.tcap; k += 64)
-      c.l_xt[k] = S.x_time[(int64_t)c.r * S.tcap + k];
-    if (lane < NS) h->arr_next[lane] = S.arr_next[(int64_t)c.r * NS + lane];
+    if (SUBW == 64) {
+      // slot finish times + transfer times into LDS (lane-strided)
+      for (int k = lane; k < S.total_slots; k += SUBW)
+        c.l_fin[k] = S.s_finish[sbase + k];
+      for (int k = lane; k < S.tcap; k += SUBW)
+        c.l_xt[k] = S.x_time[(int64_t)c.r * S.tcap + k];
+    }
+    for (int k = lane; k < NS; k += SUBW)
+      h->arr_next[k] = S.arr_next[(int64_t)c.r * NS + k];
     if (lane < nd) {
       int rd = c.r * nd + lane;
       h->dc_minf[lane] = S.dc_min_finish[rd];
@@ -949,9 +963,9 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       h->busy[lane] = S.busy[rd];
       h->n_running[lane] = S.n_running[rd];
     }
-    if (lane < nd * 2) {
-      h->q_len[lane] = S.q_len[c.r * nd * 2 + lane];
-      h->q_head[lane] = S.q_head[c.r * nd * 2 + lane];
+    for (int k = lane; k < nd * 2; k += SUBW) {
+      h->q_len[k] = S.q_len[c.r * nd * 2 + k];
+      h->q_head[k] = S.q_head[c.r * nd * 2 + k];
     }
     if (lane == 0) h->next_log = S.next_log[c.r];
   }
@@ -977,17 +991,17 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       double bw = S.wan_bw[ing * S.n_dc + d_sel];
       double xfer = bw > 0.0 ? S.payload_gb[jt] / bw : 0.0;
       int cand = INT_MAX;
-      for (int k = lane; k < S.tcap; k += 64) {
+      for (int k = lane; k < S.tcap; k += SUBW) {
         if (c.l_xt[k] >= D_INF) { cand = k; break; }
       }
 #pragma unroll
-      for (int off = 32; off > 0; off >>= 1)
+      for (int off = SUBW / 2; off > 0; off >>= 1)
         cand = min(cand, __shfl_xor(cand, off, 64));
       if (cand == INT_MAX) {
         if (lane == 0) atomicOr(&S.err[c.r], ERR_XFER_OVF);
       } else {
         int64_t at = (int64_t)c.r * S.tcap + cand;
-        for (int k = lane; k < S.obs_dim; k += 64)
+        for (int k = lane; k < S.obs_dim; k += SUBW)
           S.x_s0[at * S.obs_dim + k] = s0[k];
         if (lane == 0) {
           c.l_xt[cand] = c.now + lnet + xfer;
@@ -1032,23 +1046,23 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     // per-lane candidate: value + kind/idx
     double v = D_INF;
     int kind = -1, idx = -1;
-    // arrival streams on lanes [0, NS)
-    if (lane < NS) {
-      double t = c.hs->arr_next[lane];
-      if (t < v) { v = t; kind = 0; idx = lane; }
+    // arrival streams (strided)
+    for (int k = lane; k < NS; k += SUBW) {
+      double t = c.hs->arr_next[k];
+      if (t < v) { v = t; kind = 0; idx = k; }
     }
-    // log tick on lane NS
-    if (lane == NS) {
+    // log tick (one candidate)
+    if (lane == 0) {
       double t = c.hs->next_log;
       if (t < v) { v = t; kind = 3; idx = 0; }
     }
-    // dc min finishes on lanes [32, 32+n_dc)
-    if (lane >= 32 && lane < 32 + S.n_dc) {
-      double t = c.hs->dc_minf[lane - 32];
-      if (t < v) { v = t; kind = 2; idx = c.hs->dc_mins[lane - 32]; }
+    // dc min finishes (strided; n_dc <= 8 <= SUBW so one pass)
+    for (int k = lane; k < S.n_dc; k += SUBW) {
+      double t = c.hs->dc_minf[k];
+      if (t < v) { v = t; kind = 2; idx = c.hs->dc_mins[k]; }
     }
-    // transfers: strided over tcap (LDS mirror)
-    for (int k = lane; k < S.tcap; k += 64) {
+    // transfers (strided over the mirror)
+    for (int k = lane; k < S.tcap; k += SUBW) {
       double t = c.l_xt[k];
       if (t < v) { v = t; kind = 1; idx = k; }
     }
@@ -1147,42 +1161,65 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       if (S.trace_mode && trace_d >= 0) {
         d_sel = trace_d;
       } else if (ALGO == A_ECO_ROUTE) {
-        // all 64 lanes score: lane = d*8 + (n-1) covers (DC, n); each lane
-        // reduces over the frequency ladder, then an 8-lane subgroup min per
-        // DC and a wave argmin pick the DC (first-minimum = lowest DC index,
-        // matching the scalar scan order).
-        double score = D_INF;
-        {
-          int d = lane >> 3;
-          int n = (lane & 7) + 1;
-          if (d < S.n_dc && n <= S.max_gpj) {
+        double price = S.eco_obj == 2 ? c.price_kwh(t_min) : 0.0;
+        if (SUBW == 64) {
+          // all 64 lanes score: lane = d*8 + (n-1) covers (DC, n); each lane
+          // reduces over the frequency ladder, then an 8-lane subgroup min
+          // per DC and a wave argmin pick the DC (first-minimum = lowest DC
+          // index, matching the scalar scan order).
+          double score = D_INF;
+          {
+            int d = lane >> 3;
+            int n = (lane & 7) + 1;
+            if (d < S.n_dc && n <= S.max_gpj) {
+              const double* pcf = c.pc3(d, jt);
+              const double* tcf = c.lc3(d, jt);
+              double best = D_INF;
+              for (int q = 0; q < S.n_freq; ++q) {
+                double f = S.freq_levels[q];
+                double T = d_unit_time(n, f, tcf);
+                double E = d_job_power(n, f, pcf) * T;
+                double sc;
+                if (S.eco_obj == 1) sc = E * S.carbon[d];
+                else if (S.eco_obj == 2) sc = (E / 3.6e6) * price;
+                else sc = E;
+                if (sc < best) best = sc;
+              }
+              score = best * size;  // relative order preserved per objective
+            }
+          }
+#pragma unroll
+          for (int off = 1; off < 8; off <<= 1)
+            score = fmin(score, __shfl_xor(score, off, 64));
+          if (lane & 7) score = D_INF;
+          int dl;
+          wave_argmin_f64(score, dl);
+          d_sel = dl >> 3;
+        } else {
+          // SUBW==8: lane d scores DC d serially over the grid
+          double score = D_INF;
+          if (lane < S.n_dc) {
+            int d = lane;
             const double* pcf = c.pc3(d, jt);
             const double* tcf = c.lc3(d, jt);
             double best = D_INF;
-            for (int q = 0; q < S.n_freq; ++q) {
-              double f = S.freq_levels[q];
-              double T = d_unit_time(n, f, tcf);
-              double E = d_job_power(n, f, pcf) * T;
-              double sc;
-              if (S.eco_obj == 1) sc = E * S.carbon[d];
-              else if (S.eco_obj == 2) sc = (E / 3.6e6) * c.price_kwh(t_min);
-              else sc = E;
-              if (sc < best) best = sc;
-            }
-            score = best * size;  // relative order preserved per objective
+            for (int n = 1; n <= S.max_gpj; ++n)
+              for (int q = 0; q < S.n_freq; ++q) {
+                double f = S.freq_levels[q];
+                double T = d_unit_time(n, f, tcf);
+                double E = d_job_power(n, f, pcf) * T;
+                double sc;
+                if (S.eco_obj == 1) sc = E * S.carbon[d];
+                else if (S.eco_obj == 2) sc = (E / 3.6e6) * price;
+                else sc = E;
+                if (sc < best) best = sc;
+              }
+            score = best * size;
           }
+          int dl;
+          wave_argmin_f64(score, dl);
+          d_sel = dl & (SUBW - 1);
         }
-        // min within each DC's 8-lane subgroup
-#pragma unroll
-        for (int off = 1; off < 8; off <<= 1)
-          score = fmin(score, __shfl_xor(score, off, 64));
-        // now every lane of a DC group holds its DC's best; keep only the
-        // group leader's value for the cross-DC argmin (tie-break: lowest
-        // lane == lowest DC index)
-        if (lane & 7) score = D_INF;
-        int dl;
-        wave_argmin_f64(score, dl);
-        d_sel = dl >> 3;
       } else {
         d_sel = (int)rbelow(c.rng, (uint32_t)S.n_dc);
       }
@@ -1191,11 +1228,11 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       double xfer = bw > 0.0 ? S.payload_gb[jt] / bw : 0.0;
       // push transfer record
       int cand = INT_MAX;
-      for (int k = lane; k < S.tcap; k += 64) {
+      for (int k = lane; k < S.tcap; k += SUBW) {
         if (c.l_xt[k] >= D_INF) { cand = k; break; }
       }
 #pragma unroll
-      for (int off = 32; off > 0; off >>= 1)
+      for (int off = SUBW / 2; off > 0; off >>= 1)
         cand = min(cand, __shfl_xor(cand, off, 64));
       if (cand == INT_MAX) {
         if (lane == 0) atomicOr(&S.err[c.r], ERR_XFER_OVF);
@@ -1399,11 +1436,14 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   {
     Hot* h = c.hs;
     int nd = S.n_dc;
-    for (int k = lane; k < S.total_slots; k += 64)
-      S.s_finish[sbase + k] = c.l_fin[k];
-    for (int k = lane; k < S.tcap; k += 64)
-      S.x_time[(int64_t)c.r * S.tcap + k] = c.l_xt[k];
-    if (lane < NS) S.arr_next[(int64_t)c.r * NS + lane] = h->arr_next[lane];
+    if (SUBW == 64) {
+      for (int k = lane; k < S.total_slots; k += SUBW)
+        S.s_finish[sbase + k] = c.l_fin[k];
+      for (int k = lane; k < S.tcap; k += SUBW)
+        S.x_time[(int64_t)c.r * S.tcap + k] = c.l_xt[k];
+    }
+    for (int k = lane; k < NS; k += SUBW)
+      S.arr_next[(int64_t)c.r * NS + k] = h->arr_next[k];
     if (lane < nd) {
       int rd = c.r * nd + lane;
       S.dc_min_finish[rd] = h->dc_minf[lane];
@@ -1418,9 +1458,9 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       S.busy[rd] = h->busy[lane];
       S.n_running[rd] = h->n_running[lane];
     }
-    if (lane < nd * 2) {
-      S.q_len[c.r * nd * 2 + lane] = h->q_len[lane];
-      S.q_head[c.r * nd * 2 + lane] = h->q_head[lane];
+    for (int k = lane; k < nd * 2; k += SUBW) {
+      S.q_len[c.r * nd * 2 + k] = h->q_len[k];
+      S.q_head[c.r * nd * 2 + k] = h->q_head[k];
     }
   }
   if (lane == 0) {
@@ -1565,14 +1605,16 @@ class BatchedSimHip {
 
   // launch one advance chunk; returns immediately (stream-async)
   void advance(double t_target, int64_t max_ev) {
-    int waves_per_block = WAVES_PER_BLOCK;
-    int blocks = (S_.n_rep + waves_per_block - 1) / waves_per_block;
-    dim3 grid(blocks), block(64 * waves_per_block);
-    // dynamic LDS: per wave, Hot + s_finish mirror + x_time mirror
+    int rpb = REPLICAS_PER_BLOCK;
+    int blocks = (S_.n_rep + rpb - 1) / rpb;
+    dim3 grid(blocks), block(THREADS_PER_BLOCK);
+    // dynamic LDS: per replica, Hot (+ s_finish/x_time mirrors at SUBW==64)
     size_t hot_sz = (sizeof(Hot) + 15) & ~size_t(15);
-    size_t shmem = waves_per_block *
-        (hot_sz + (size_t)S_.total_slots * sizeof(double) +
-         (size_t)S_.tcap * sizeof(double));
+    size_t per_rep = hot_sz;
+    if (SUBW == 64)
+      per_rep += (size_t)S_.total_slots * sizeof(double) +
+                 (size_t)S_.tcap * sizeof(double);
+    size_t shmem = rpb * per_rep;
     if (shmem > 64 * 1024)
       throw std::runtime_error(
           "scenario too large for the LDS-mirrored engine (total_slots + "

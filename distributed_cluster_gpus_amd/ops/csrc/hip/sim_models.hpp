@@ -9,6 +9,7 @@
 // (SURVEY §2 row 8).
 #pragma once
 #include <hip/hip_runtime.h>
+#include <limits.h>
 
 namespace dcg {
 
@@ -28,21 +29,35 @@ __device__ __forceinline__ double d_unit_time(int n, double f, const double* c3)
   return (c3[0] + c3[1] / f + c3[2] * n) / n;
 }
 
-// ---- wavefront reductions (64-wide; __shfl_xor over all 64 lanes) ----
+// ---- subwave reductions ----
+// DCG_SUBWAVE lanes cooperate on one replica (64 = classic wave-per-replica;
+// 8 = eight replicas per wavefront, amortizing the wave-uniform scalar work
+// across subgroups).  xor-butterfly offsets < SUBW stay inside an aligned
+// subgroup, so the same shfl_xor code serves both widths.
+#ifndef DCG_SUBWAVE
+#define DCG_SUBWAVE 64
+#endif
+constexpr int SUBW = DCG_SUBWAVE;
+static_assert(SUBW == 64 || SUBW == 8, "supported subwave widths: 64, 8");
+
+__device__ __forceinline__ int sub_lane() { return threadIdx.x & (SUBW - 1); }
+
 __device__ __forceinline__ double wave_min_f64(double v) {
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1)
+  for (int off = SUBW / 2; off > 0; off >>= 1)
     v = fmin(v, __shfl_xor(v, off, 64));
   return v;
 }
 
-// argmin with lowest-lane tie-break: returns min value; *lane_out = winner.
+// argmin with lowest-absolute-lane tie-break within the subgroup: returns the
+// min value; lane_out = the winning lane's ABSOLUTE in-wave index (valid as
+// a __shfl source for the subgroup).
 __device__ __forceinline__ double wave_argmin_f64(double v, int& lane_out) {
   int lane = threadIdx.x & 63;
   double bv = v;
   int bl = lane;
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) {
+  for (int off = SUBW / 2; off > 0; off >>= 1) {
     double ov = __shfl_xor(bv, off, 64);
     int ol = __shfl_xor(bl, off, 64);
     if (ov < bv || (ov == bv && ol < bl)) { bv = ov; bl = ol; }
@@ -51,9 +66,26 @@ __device__ __forceinline__ double wave_argmin_f64(double v, int& lane_out) {
   return bv;
 }
 
-// 64-candidate (n, f) grid argmin.  lane = (n-1)*n_freq + f_idx reproduces the
+// generic (value, index) argmin: FIRST minimum in candidate-index order,
+// matching the scalar engines' scan-order tie-break exactly.
+__device__ __forceinline__ void wave_argmin_idx_f64(double v, int idx,
+                                                    double& v_out, int& i_out) {
+  double bv = v;
+  int bi = idx;
+#pragma unroll
+  for (int off = SUBW / 2; off > 0; off >>= 1) {
+    double ov = __shfl_xor(bv, off, 64);
+    int oi = __shfl_xor(bi, off, 64);
+    if (ov < bv || (ov == bv && oi < bi)) { bv = ov; bi = oi; }
+  }
+  v_out = bv;
+  i_out = bi;
+}
+
+// (n, f) grid argmin over n_max*n_freq candidates, candidate-strided over the
+// subgroup's lanes; candidate index ci = (n-1)*n_freq + f_idx reproduces the
 // scalar scan order (n-major, frequency-minor, first-minimum tie-break) of
-// policies/gridsearch.py::best_nf_grid.  Lanes beyond the grid contribute +inf.
+// policies/gridsearch.py::best_nf_grid.
 // objective: 0 energy, 1 carbon (score=E*ci), 2 cost (score=E/3.6e6*price).
 struct GridPick { int n; double f, T, P, E; bool found; };
 
@@ -61,47 +93,55 @@ __device__ __forceinline__ GridPick wave_grid_argmin(
     const double* pc3, const double* lc3, const double* freq_levels,
     int n_freq, int n_max, int objective, double ci, double price,
     bool has_ddl, double ddl) {
-  int lane = threadIdx.x & 63;
-  int n = lane / n_freq + 1;
-  int fi = lane % n_freq;
-  double score = D_INF, T = 0, P = 0, E = 0, f = 0;
-  if (lane < n_max * n_freq) {
-    f = freq_levels[fi];
-    T = d_unit_time(n, f, lc3);
-    P = d_job_power(n, f, pc3);
-    E = P * T;
-    if (!(has_ddl && T > ddl)) {
-      score = E;
-      if (objective == 1) score = E * ci;
-      else if (objective == 2) score = (E / 3.6e6) * price;
-    }
+  int lane = sub_lane();
+  int n_cand = n_max * n_freq;
+  double best_sc = D_INF;
+  int best_ci = INT_MAX;
+  for (int cand = lane; cand < n_cand; cand += SUBW) {
+    int n = cand / n_freq + 1;
+    double f = freq_levels[cand % n_freq];
+    double T = d_unit_time(n, f, lc3);
+    double E = d_job_power(n, f, pc3) * T;
+    if (has_ddl && T > ddl) continue;
+    double sc = E;
+    if (objective == 1) sc = E * ci;
+    else if (objective == 2) sc = (E / 3.6e6) * price;
+    if (sc < best_sc) { best_sc = sc; best_ci = cand; }
   }
-  int wl;
-  double best = wave_argmin_f64(score, wl);
+  double v;
+  int wi;
+  wave_argmin_idx_f64(best_sc, best_ci, v, wi);
   GridPick out;
-  out.found = best < D_INF;
-  out.n = __shfl(n, wl, 64);
-  out.f = __shfl(f, wl, 64);
-  out.T = __shfl(T, wl, 64);
-  out.P = __shfl(P, wl, 64);
-  out.E = __shfl(E, wl, 64);
+  out.found = v < D_INF;
+  if (!out.found) {
+    out.n = 1; out.f = 0; out.T = 0; out.P = 0; out.E = 0;
+    return out;
+  }
+  out.n = wi / n_freq + 1;
+  out.f = freq_levels[wi % n_freq];
+  out.T = d_unit_time(out.n, out.f, lc3);
+  out.P = d_job_power(out.n, out.f, pc3);
+  out.E = out.P * out.T;
   return out;
 }
 
-// energy-argmin over frequencies at fixed n (best_energy_freq): lane-parallel
-// over n_freq lanes, first-minimum tie-break.
+// energy-argmin over frequencies at fixed n (best_energy_freq),
+// candidate-strided, first-minimum tie-break.
 __device__ __forceinline__ double wave_energy_freq(
     const double* pc3, const double* lc3, const double* freq_levels,
     int n_freq, int n) {
-  int lane = threadIdx.x & 63;
-  double score = D_INF, f = 0;
-  if (lane < n_freq) {
-    f = freq_levels[lane];
-    score = d_job_power(n, f, pc3) * d_unit_time(n, f, lc3);
+  int lane = sub_lane();
+  double best_sc = D_INF;
+  int best_k = INT_MAX;
+  for (int k = lane; k < n_freq; k += SUBW) {
+    double f = freq_levels[k];
+    double sc = d_job_power(n, f, pc3) * d_unit_time(n, f, lc3);
+    if (sc < best_sc) { best_sc = sc; best_k = k; }
   }
-  int wl;
-  wave_argmin_f64(score, wl);
-  return __shfl(f, wl, 64);
+  double v;
+  int wk;
+  wave_argmin_idx_f64(best_sc, best_k, v, wk);
+  return freq_levels[wk];
 }
 
 }  // namespace dcg

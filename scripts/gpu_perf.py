@@ -13,13 +13,13 @@ from distributed_cluster_gpus_amd.configs.paper import build_arrivals, paper_sce
 from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
 
 
-def run_one(replicas, duration=1200.0, algo="default_policy", qcap=24576):
+def run_one(replicas, duration=1200.0, algo="default_policy", qcap=24576, subwave=64):
     sc = paper_scenario()
     inf, trn = build_arrivals()
     eng = BatchedEngine(sc, inf, trn, algo=algo, replicas=replicas,
                         duration=duration, log_interval=5.0, out_dir=None,
                         seed=123, enable_logs=False, qcap=qcap,
-                        events_per_launch=100000)
+                        events_per_launch=100000, subwave=subwave)
     t0 = time.perf_counter()
     st = eng.run()
     wall = time.perf_counter() - t0
@@ -36,6 +36,11 @@ def run_one(replicas, duration=1200.0, algo="default_policy", qcap=24576):
 
 
 if __name__ == "__main__":
-    reps = [int(x) for x in (sys.argv[1:] or ["512", "2048", "8192"])]
+    args = sys.argv[1:]
+    sw = 64
+    if args and args[0] == "--mw":
+        sw = 8
+        args = args[1:]
+    reps = [int(x) for x in (args or ["512", "2048", "8192"])]
     for r in reps:
-        print(json.dumps(run_one(r)), flush=True)
+        print(json.dumps(run_one(r, subwave=sw)), flush=True)
