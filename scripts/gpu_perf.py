@@ -4,6 +4,8 @@ on the BASELINE config (default_policy, 1200 s, paper topology) and report
 aggregate events/sec for several replica counts."""
 import json
 import sys
+
+import torch
 import time
 import os, sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
@@ -44,3 +46,6 @@ if __name__ == "__main__":
     reps = [int(x) for x in (args or ["512", "2048", "8192"])]
     for r in reps:
         print(json.dumps(run_one(r, subwave=sw)), flush=True)
+        import gc
+        gc.collect()
+        torch.cuda.empty_cache()
