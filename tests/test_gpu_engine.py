@@ -442,3 +442,18 @@ def test_single_replica_exact_trajectory_parity(tmp_path, algo, kw,
         a, b = m[f"{col}_o"], m[f"{col}_g"]
         d = ((a - b).abs() - (1.01 * quantum + rtol * a.abs())).max()
         assert d <= 0, f"cluster {col} beyond print+fp tolerance by {d}"
+
+
+@needs_gpu
+def test_subwave_variants_bit_identical():
+    """The 8-replicas-per-wave engine (subwave=8) must produce bit-identical
+    trajectories to the default wave-per-replica engine — same Philox streams,
+    same f64 math, different lane geometry."""
+    e64 = make_engine(replicas=64, duration=90.0)
+    e64.run()
+    e8 = make_engine(replicas=64, duration=90.0, subwave=8)
+    e8.run()
+    for key in ("ev_count", "jobs_done", "jobs_done_inf"):
+        assert torch.equal(e64.t[key].cpu(), e8.t[key].cpu()), key
+    assert torch.equal(e64.t["energy_j"].cpu(), e8.t["energy_j"].cpu())
+    assert torch.equal(e64.t["sum_lat"].cpu(), e8.t["sum_lat"].cpu())
