@@ -1660,7 +1660,9 @@ class BatchedSimHip {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "Batched MI355X (gfx950) Monte-Carlo replica engine";
-  py::class_<dcg::BatchedSimHip>(m, "BatchedSimHip")
+  // module_local: the 64-wide and 8-wide extensions register the
+  // same C++ type in one process
+  py::class_<dcg::BatchedSimHip>(m, "BatchedSimHip", py::module_local())
       .def(py::init<py::dict, py::dict>())
       .def("advance", &dcg::BatchedSimHip::advance,
            py::arg("t_target"), py::arg("max_events"));
