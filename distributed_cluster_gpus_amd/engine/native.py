@@ -29,6 +29,7 @@ class NativeEngine:
                  power_cap: float = 0.0, control_interval: float = 5.0,
                  elastic_scaling: bool = False, eco_objective: str = "energy",
                  num_fixed_gpus: int = 1, fixed_freq: Optional[float] = None,
+                 use_control_interval: bool = False,
                  logger=None, show_progress: bool = False, **rl_kwargs):
         if algo not in ALGOS:
             raise ValueError(f"unknown algo {algo!r}")
@@ -40,6 +41,7 @@ class NativeEngine:
                 scenario, arrival_inf, arrival_trn, algo=algo,
                 duration=duration, log_interval=log_interval, out_dir=out_dir,
                 seed=seed, power_cap=power_cap, control_interval=control_interval,
+                use_control_interval=use_control_interval,
                 elastic_scaling=elastic_scaling, eco_objective=eco_objective,
                 num_fixed_gpus=num_fixed_gpus, fixed_freq=fixed_freq,
                 logger=logger, show_progress=show_progress, **rl_kwargs)

@@ -56,6 +56,7 @@ class OracleEngine:
                  duration: float = 3600.0, log_interval: float = 10.0,
                  out_dir: Optional[str] = None, seed: int = 42,
                  power_cap: float = 0.0, control_interval: float = 5.0,
+                 use_control_interval: bool = False,
                  elastic_scaling: bool = False, eco_objective: str = "energy",
                  num_fixed_gpus: int = 1, fixed_freq: Optional[float] = None,
                  sla_p99_ms: float = 500.0, energy_budget_j: Optional[float] = None,
@@ -73,6 +74,10 @@ class OracleEngine:
         self.log_interval = float(log_interval)
         self.power_cap = float(power_cap)
         self.control_interval = float(control_interval)
+        # Reference parity: the cap controller fires on every LOG tick and
+        # --control-interval is parsed but unused.
+        # use_control_interval=True opts into a dedicated control cadence.
+        self.use_control_interval = bool(use_control_interval)
         self.eco_objective = eco_objective
         self.num_fixed_gpus = int(num_fixed_gpus)
         self.fixed_freq = fixed_freq
@@ -154,6 +159,8 @@ class OracleEngine:
             self._schedule(self.now + self.arr_trn.next_interarrival(self.now, self.rng),
                            "arrival_trn", {"ing": ing_name})
         self._schedule(self.now + self.log_interval, "log", {"interval": self.log_interval})
+        if self.use_control_interval and self.power_cap > 0:
+            self._schedule(self.now + self.control_interval, "control", {})
 
     # ---------- event plumbing ----------
     def _schedule(self, t: float, etype: str, payload: dict):
@@ -234,8 +241,12 @@ class OracleEngine:
                 if payload.get("gen") != job.ev_gen:
                     continue  # stale finish event (lazy invalidation)
                 self._on_job_finish(payload["dc"], payload["jid"])
-            elif etype == "log":
+            elif etype == "control":
                 self._control()
+                self._schedule(self.now + self.control_interval, "control", {})
+            elif etype == "log":
+                if not self.use_control_interval:
+                    self._control()
                 self._on_log(payload["interval"])
             else:
                 raise RuntimeError(f"Unknown event {etype}")

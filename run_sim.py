@@ -59,6 +59,10 @@ def parse_args(argv=None):
                    help="Total power cap (W); only cap_uniform/cap_greedy, <=0 = off.")
     p.add_argument("--control-interval", type=float, default=5.0,
                    help="Controller cadence (seconds).")
+    p.add_argument("--use-control-interval", action="store_true", default=False,
+                   help="Actually fire the cap controller at --control-interval "
+                        "(the reference parses the flag but fires at "
+                        "log-interval; default keeps that parity).")
     p.add_argument("--eco-objective", type=str, default="energy",
                    choices=["energy", "carbon", "cost"])
     # debug params
@@ -134,6 +138,7 @@ def main(argv=None):
         out_dir=out_dir, seed=args.seed, power_cap=args.power_cap,
         control_interval=args.control_interval,
         elastic_scaling=args.elastic_scaling, eco_objective=args.eco_objective,
+        use_control_interval=args.use_control_interval,
         num_fixed_gpus=args.num_fixed_gpus, fixed_freq=args.fixed_freq,
         sla_p99_ms=args.sla_p99_ms, energy_budget_j=args.energy_budget_j,
         rl_device=rl_device, rl_batch=args.upgr_batch,
