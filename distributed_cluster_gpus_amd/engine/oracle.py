@@ -99,6 +99,18 @@ class OracleEngine:
 
         self.dcs: Dict[str, DataCenterState] = scenario.make_dc_states()
         self._dc_names = list(self.dcs.keys())
+        self._dc_idx = {n: i for i, n in enumerate(self._dc_names)}
+        self._ing_idx = {n: i for i, n in enumerate(scenario.ingress_names)}
+        # coefficient cache: (dc, jtype) -> (PowerCoeffs, LatencyCoeffs, raw
+        # power triple); identical values to the scenario tables, avoids
+        # rebuilding dataclasses in the per-event hot loop
+        from ..models.coeffs import LatencyCoeffs, PowerCoeffs
+        self._coeff_cache = {}
+        for d, name in enumerate(self._dc_names):
+            for j, jname in enumerate(("inference", "training")):
+                pc = PowerCoeffs(*scenario.power_coeffs[d, j, :])
+                tc = LatencyCoeffs(*scenario.latency_coeffs[d, j, :])
+                self._coeff_cache[(name, jname)] = (pc, tc)
         self.event_q: List[Tuple[float, int, str, dict]] = []
         self._seq = itertools.count()
         self._jid = itertools.count(1)
@@ -172,23 +184,20 @@ class OracleEngine:
     def _dc_power_w(self, dc: DataCenterState) -> float:
         """Paper power model: sum of per-job n*P(f_used) + sleeping/idle floor
         (reference _estimate_dc_power, :168-179)."""
-        d = self._dc_names.index(dc.name)
+        cache = self._coeff_cache
+        name = dc.name
         p_active = 0.0
         for job, g in dc.running_jobs.values():
-            j = 0 if job.jtype == "inference" else 1
-            a, b, c = self.sc.power_coeffs[d, j, :]
+            pc = cache[(name, job.jtype)][0]
             f = max(0.0, job.f_used)
-            p_active += max(0, int(g)) * (a * f ** 3 + b * f + c)
+            p_active += max(0, int(g)) * (pc.alpha_p * f ** 3 + pc.beta_p * f
+                                          + pc.gamma_p)
         idle = dc.total_gpus - dc.busy_gpus
         p_idle = idle * (dc.p_sleep if dc.power_gating else dc.p_idle)
         return p_active + p_idle
 
     def _coeffs(self, dc_name: str, jtype: str):
-        from ..models.coeffs import LatencyCoeffs, PowerCoeffs
-        d = self._dc_names.index(dc_name)
-        j = 0 if jtype == "inference" else 1
-        return (PowerCoeffs(*self.sc.power_coeffs[d, j, :]),
-                LatencyCoeffs(*self.sc.latency_coeffs[d, j, :]))
+        return self._coeff_cache[(dc_name, jtype)]
 
     # ---------- run loop ----------
     def run(self):
@@ -283,8 +292,8 @@ class OracleEngine:
         """(Lnet_s, bottleneck_gbps, cost_per_gb, transfer_s) from precomputed
         all-pairs tables (graph is static; reference runs Dijkstra per arrival,
         :482-496 — identical numbers)."""
-        i = self.sc.ingress_names.index(ing_name)
-        d = self._dc_names.index(dc_name)
+        i = self._ing_idx[ing_name]
+        d = self._dc_idx[dc_name]
         lnet = float(self.sc.wan_latency_s[i][d])
         bw = float(self.sc.wan_bottleneck_gbps[i][d])
         cost = float(self.sc.wan_cost_per_gb[i][d])
@@ -800,7 +809,6 @@ class OracleEngine:
     def _cap_greedy(self, deficit: float, totalP: float):
         """Per-job atom-based capping: apply down-atoms cheapest-rho-first with
         exact power re-estimation after each (reference :248-315)."""
-        from ..models.coeffs import LatencyCoeffs, PowerCoeffs
         guard = 10000
         while deficit > 1e-6 and guard > 0:
             guard -= 1
@@ -810,17 +818,14 @@ class OracleEngine:
                 if not levels:
                     continue
                 f_min = min(levels)
-                d = self._dc_names.index(dc.name)
                 for job, g in list(dc.running_jobs.values()):
                     cur_f = job.f_used or dc.current_freq
                     if cur_f <= f_min + 1e-12:
                         continue
-                    j = 0 if job.jtype == "inference" else 1
+                    pc, tc = self._coeff_cache[(dc.name, job.jtype)]
                     tasks.append(RunningTask(
                         job_id=job.jid, dc_name=dc.name, n=g, f=cur_f,
-                        freq_levels=levels,
-                        pc=PowerCoeffs(*self.sc.power_coeffs[d, j, :]),
-                        tc=LatencyCoeffs(*self.sc.latency_coeffs[d, j, :])))
+                        freq_levels=levels, pc=pc, tc=tc))
             if not tasks:
                 break
             _, down_atoms = aggregate_atoms(tasks)

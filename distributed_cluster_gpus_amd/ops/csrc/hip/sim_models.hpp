@@ -2,11 +2,12 @@
 //
 // The DVFS power polynomial and T(n,f) latency model are fused into every
 // step kernel as inlined device functions ("HIP device
-// function, fused").  f64 forms are used on the simulation path (energy
-// integrals and event times need the precision); the wave-parallel grid
-// search evaluates all 64 (n,f) candidates with one lane each — the 8x8
-// candidate grid maps exactly onto CDNA4's 64-wide wavefront
-// (SURVEY §2 row 8).
+// function, fused").  f64 forms are used on the simulation path (event times
+// must be BITWISE-equal to the scalar engines; energy integrals need the
+// precision).  The (n,f) grid search is candidate-strided over the replica's
+// subgroup: at the default 64-lane width the 8x8 candidate grid maps exactly
+// one-candidate-per-lane onto a CDNA4 wavefront; the
+// 8-lane multi-replica build strides 8 candidates per lane.
 #pragma once
 #include <hip/hip_runtime.h>
 #include <limits.h>
