@@ -155,6 +155,9 @@ class BatchedEngine:
         t["s_gpus"] = torch.zeros((R, total_slots), **i16)
         t["s_jtype"] = torch.zeros((R, total_slots), **i8)
         t["s_ing"] = torch.zeros((R, total_slots), **i8)
+        t["s_done"] = torch.zeros((R, total_slots), **f64)
+        t["s_pcount"] = torch.zeros((R, total_slots),
+                                    dtype=torch.uint8, device=dev)
         t["x_time"] = torch.full((R, tcap), INF, **f64)
         t["x_size"] = torch.zeros((R, tcap), **f64)
         t["x_netlat"] = torch.zeros((R, tcap), **f32)
@@ -189,7 +192,7 @@ class BatchedEngine:
         t["cl_count"] = torch.zeros(1, **i32)
         t["cl_rows"] = torch.zeros((cl_cap, 16), **f64)
         t["jl_count"] = torch.zeros(1, **i32)
-        t["jl_rows"] = torch.zeros((jl_cap, 10), **f64)
+        t["jl_rows"] = torch.zeros((jl_cap, 11), **f64)
 
         # arrival-trace replay mode (exact single-replica parity testing):
         # arrivals come from a recorded (time, size) FIFO per stream
@@ -259,6 +262,24 @@ class BatchedEngine:
             t["x_mdc"] = torch.zeros((R, tcap), **u8)
             t["x_mg"] = torch.zeros((R, tcap), **u8)
             t["x_has_rl"] = torch.zeros((R, tcap), **u8)
+            # elastic-scaling preempted-job pool
+            pp_cap = 64
+            self._pp_cap = pp_cap
+            u8_ = dict(dtype=torch.uint8, device=dev)
+            t["pp_count"] = torch.zeros(R, **i32)
+            t["pp_cursor"] = torch.zeros(R, **i32)
+            t["pp_size"] = torch.zeros((R, pp_cap), **f64)
+            t["pp_done"] = torch.zeros((R, pp_cap), **f64)
+            t["pp_netlat"] = torch.zeros((R, pp_cap), **f32)
+            t["pp_jid"] = torch.zeros((R, pp_cap), **i32)
+            t["pp_ing"] = torch.zeros((R, pp_cap), **u8_)
+            t["pp_dc"] = torch.zeros((R, pp_cap), **u8_)
+            t["pp_pcount"] = torch.zeros((R, pp_cap), **u8_)
+            t["pp_s0"] = torch.zeros((R, pp_cap, obs_dim), **f32)
+            t["pp_adc"] = torch.zeros((R, pp_cap), **u8_)
+            t["pp_ag"] = torch.zeros((R, pp_cap), **u8_)
+            t["pp_nrew"] = torch.zeros((R, pp_cap), **u8_)
+            t["pp_has_rl"] = torch.zeros((R, pp_cap), **u8_)
             t["lat_hist"] = torch.zeros((R, 2, 64), **i32)
             t["lat_count"] = torch.zeros((R, 2), **i64)
             t["lat_sum"] = torch.zeros((R, 2), **f64)
@@ -329,6 +350,8 @@ class BatchedEngine:
             "cl_cap": cl_cap, "jl_cap": jl_cap,
             "obs_dim": obs_dim, "sla_p99_ms": float(sla_p99_ms),
             "tr_cap": int(tr_cap) if self.is_rl else 0,
+            "elastic": int(bool(elastic_scaling) and self.is_rl),
+            "pp_cap": getattr(self, "_pp_cap", 0),
             "trace_mode": int(self.trace_mode),
             "trace_cap": getattr(self, "trace_cap", 0),
         }
@@ -584,7 +607,8 @@ class BatchedEngine:
         from ..models.coeffs import LatencyCoeffs, PowerCoeffs
         from ..policies.gridsearch import energy_tuple
         for row in jrows:
-            jid, ing, jt, size, d, fused, n, netlat, start, finish = row
+            (jid, ing, jt, size, d, fused, n, netlat, start, finish,
+             pcount) = row
             d = int(d)
             jt = int(jt)
             pC = PowerCoeffs(*self.sc.power_coeffs[d, jt, :])
@@ -593,5 +617,6 @@ class BatchedEngine:
             jw.row(int(jid), self.sc.ingress_names[int(ing)],
                    "inference" if jt == 0 else "training", float(size),
                    self.sc.dc_names[d], float(fused), int(n), float(netlat),
-                   float(start), float(finish), 0, T_pred, P_pred, E_pred)
+                   float(start), float(finish), int(pcount),
+                   T_pred, P_pred, E_pred)
         jw.close()

@@ -44,7 +44,8 @@ enum Algo { A_DEFAULT = 0, A_CAP_UNIFORM, A_CAP_GREEDY, A_JOINT_NF, A_BANDIT,
             A_CARBON_COST, A_ECO_ROUTE, A_DEBUG, A_CHSAC };
 
 // CHSAC-AF action-request state machine kinds
-enum PendKind : int { PEND_NONE = 0, PEND_ARRIVAL = 1, PEND_DRAIN = 2 };
+enum PendKind : int { PEND_NONE = 0, PEND_ARRIVAL = 1, PEND_DRAIN = 2,
+                      PEND_REALLOC = 3 };
 // request flags
 enum ReqFlag : int { REQ_IDLE = 0, REQ_PENDING = 1, REQ_READY = 2 };
 
@@ -115,6 +116,8 @@ struct EngineDesc {
   short* s_gpus;                  // 0 = empty
   char* s_jtype;
   char* s_ing;
+  double* s_done;                 // units completed before this (re)start
+  unsigned char* s_pcount;        // preemptions experienced
   // in-flight transfers [r][tcap]
   double* x_time;                 // INF = empty
   double* x_size;
@@ -195,6 +198,23 @@ struct EngineDesc {
   unsigned char* x_mdc;           // [r][tcap]
   unsigned char* x_mg;            // [r][tcap]
   unsigned char* x_has_rl;        // [r][tcap]
+  // elastic scaling (chsac only): preempted-training-job pool per replica
+  int elastic;                    // 0 off, 1 on
+  int pp_cap;                     // pool capacity
+  int* pp_count;                  // [r] entries in pool
+  int* pp_cursor;                 // [r] next entry to reallocate
+  double* pp_size;                // [r][cap] original size
+  double* pp_done;                // [r][cap] units completed at preemption
+  float* pp_netlat;               // [r][cap]
+  int* pp_jid;                    // [r][cap]
+  unsigned char* pp_ing;          // [r][cap]
+  unsigned char* pp_dc;           // [r][cap]
+  unsigned char* pp_pcount;       // [r][cap] job's preempt count
+  float* pp_s0;                   // [r][cap][obs_dim] carried RL trace
+  unsigned char* pp_adc;          // [r][cap]
+  unsigned char* pp_ag;           // [r][cap]
+  unsigned char* pp_nrew;         // [r][cap]
+  unsigned char* pp_has_rl;       // [r][cap]
   // latency histograms per (r, jtype): counts in log10 bins
   int* lat_hist;                  // [r][2][LAT_BINS]
   long long* lat_count;           // [r][2]
@@ -336,6 +356,8 @@ __device__ void start_job(Ctx& c, int d, int jt, double size, float netlat,
     S.s_gpus[base + cand] = (short)n;
     S.s_jtype[base + cand] = (char)jt;
     S.s_ing[base + cand] = (char)ing;
+    S.s_done[base + cand] = 0.0;
+    S.s_pcount[base + cand] = 0;
     c.hs->busy[d] += n;
     c.hs->n_running[d] += 1;
     c.hs->p_active[d] += d_job_power(n, f, c.pc3(d, jt));
@@ -594,16 +616,16 @@ __device__ void emit_cluster_rows(Ctx& c, double now) {
 
 __device__ void emit_job_row(Ctx& c, int d, int jt, int jid, int ing,
                              double size, double fused, int n, float netlat,
-                             double start, double finish) {
+                             double start, double finish, int pcount) {
   const EngineDesc& S = *c.S;
   if (c.r != S.log_replica) return;
   if (c.lane == 0) {
     int idx = *S.jl_count;
     if (idx < S.jl_cap) {
-      double* row = &S.jl_rows[(int64_t)idx * 10];
+      double* row = &S.jl_rows[(int64_t)idx * 11];
       row[0] = jid; row[1] = ing; row[2] = jt; row[3] = size; row[4] = d;
       row[5] = fused; row[6] = n; row[7] = netlat; row[8] = start;
-      row[9] = finish;
+      row[9] = finish; row[10] = pcount;
       *S.jl_count = idx + 1;
     } else {
       atomicOr(&S.err[c.r], ERR_LOG_OVF);
@@ -805,7 +827,9 @@ __device__ double rl_energy_freq(Ctx& c, int d, int jt, int n) {
 __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
                              int jid, int ing, int n, double f, double now,
                              const float* s0, int a_dc, int a_g,
-                             int mdc, int mg, int n_rew) {
+                             int mdc, int mg, int n_rew,
+                             double units_done = 0.0, int pcount = 0,
+                             int has_rl = 1) {
   const EngineDesc& S = *c.S;
   int64_t base = (int64_t)c.r * S.total_slots;
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
@@ -821,7 +845,10 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
     return;
   }
   double T = d_unit_time(n, f, c.lc3(d, jt));
-  double finish = now + (double)size * T;
+  // resume semantics: remaining units at the new (n, f)
+  // (reference _resume_preempted_job, :362-387)
+  double units_left = fmax(0.0, size - units_done);
+  double finish = now + units_left * T;
   // copy s0 trace (lane-parallel over obs_dim)
   for (int k = c.lane; k < S.obs_dim; k += SUBW)
     S.slot_s0[(base + cand) * S.obs_dim + k] = s0[k];
@@ -835,11 +862,13 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
     S.s_gpus[base + cand] = (short)n;
     S.s_jtype[base + cand] = (char)jt;
     S.s_ing[base + cand] = (char)ing;
+    S.s_done[base + cand] = units_done;
+    S.s_pcount[base + cand] = (unsigned char)pcount;
     S.slot_adc[base + cand] = (unsigned char)a_dc;
     S.slot_ag[base + cand] = (unsigned char)a_g;
     S.slot_mdc[base + cand] = (unsigned char)mdc;
     S.slot_mg[base + cand] = (unsigned char)mg;
-    S.slot_has_rl[base + cand] = 1;
+    S.slot_has_rl[base + cand] = (unsigned char)has_rl;
     S.slot_nrew[base + cand] = (unsigned char)max(1, n_rew);
     c.hs->busy[d] += n;
     c.hs->n_running[d] += 1;
@@ -888,6 +917,97 @@ __device__ void rl_emit_transition(Ctx& c, const float* s0, int a_dc, int a_g,
   store_fence();
 }
 
+// ---- elastic scaling (chsac): preempt all training jobs of a DC into the
+// replica's pool, then reallocate them one by one with fresh RL actions
+// (reference _preempt_all_training_jobs :396-409 + _rl_reallocate_training_jobs
+// :498-534; a failed resume re-queues instead of stranding — oracle fix) ----
+__device__ int rl_elastic_preempt_all(Ctx& c, int d, double now) {
+  const EngineDesc& S = *c.S;
+  int64_t base = (int64_t)c.r * S.total_slots;
+  int lo = S.slot_off[d], hi = S.slot_off[d + 1];
+  int count = 0;
+  for (int k = lo; k < hi; ++k) {   // uniform serial walk (rare event)
+    if (S.s_gpus[base + k] == 0 || S.s_jtype[base + k] != 1) continue;
+    if (count >= S.pp_cap) {
+      if (c.lane == 0) atomicOr(&S.err[c.r], ERR_SLOT_OVF);
+      break;
+    }
+    int n = S.s_gpus[base + k];
+    double f = S.s_fused[base + k];
+    double T = d_unit_time(n, f, c.lc3(d, 1));
+    double done = S.s_done[base + k] +
+                  fmax(0.0, now - S.s_start[base + k]) / fmax(T, 1e-300);
+    double size = S.s_size[base + k];
+    done = fmin(size, done);
+    int64_t pb = (int64_t)c.r * S.pp_cap + count;
+    if (c.lane == 0) {
+      S.pp_size[pb] = size;
+      S.pp_done[pb] = done;
+      S.pp_netlat[pb] = S.s_netlat[base + k];
+      S.pp_jid[pb] = S.s_jid[base + k];
+      S.pp_ing[pb] = (unsigned char)S.s_ing[base + k];
+      S.pp_dc[pb] = (unsigned char)d;
+      S.pp_pcount[pb] = (unsigned char)(S.s_pcount[base + k] + 1);
+      S.pp_adc[pb] = S.slot_adc[base + k];
+      S.pp_ag[pb] = S.slot_ag[base + k];
+      S.pp_nrew[pb] = S.slot_nrew[base + k];
+      S.pp_has_rl[pb] = S.slot_has_rl[base + k];
+      // free the slot + caches
+      S.s_gpus[base + k] = 0;
+      c.l_fin[k] = D_INF;
+      c.hs->busy[d] -= n;
+      c.hs->n_running[d] -= 1;
+      c.hs->p_active[d] -= d_job_power(n, f, c.pc3(d, 1));
+      c.hs->sum_tpt[d] -= 1.0 / T;
+    }
+    store_fence();
+    // carry the RL trace (lane-parallel)
+    for (int q = c.lane; q < S.obs_dim; q += SUBW)
+      S.pp_s0[pb * S.obs_dim + q] = S.slot_s0[(base + k) * S.obs_dim + q];
+    store_fence();
+    ++count;
+  }
+  if (c.lane == 0) {
+    S.pp_count[c.r] = count;
+    S.pp_cursor[c.r] = 0;
+  }
+  store_fence();
+  rescan_dc_min(c, d);
+  return count;
+}
+
+// chsac drains AT MOST ONE queued job per finish via a fresh policy action
+// (reference :849-890); returns true if a request was issued (pause)
+__device__ bool rl_try_drain_request(Ctx& c, int d, double now) {
+  const EngineDesc& S = *c.S;
+  double qsize;
+  float qnetlat;
+  int qjid, qing;
+  bool popped = false;
+  int from_inf = 0;
+  if (c.free_gpus(d) > 0) {
+    if (S.inf_priority && queue_pop(c, d, 0, qsize, qnetlat, qjid, qing)) {
+      popped = true;
+      from_inf = 1;
+    } else if (queue_pop(c, d, 1, qsize, qnetlat, qjid, qing)) {
+      popped = true;
+    }
+  }
+  if (popped)
+    rl_request(c, PEND_DRAIN, now, from_inf ? 0 : 1, qing, qsize, qnetlat,
+               qjid, d, from_inf);
+  return popped;
+}
+
+// request the RL action for the pool entry at the cursor
+__device__ void rl_request_realloc(Ctx& c, double now) {
+  const EngineDesc& S = *c.S;
+  int cur = S.pp_cursor[c.r];
+  int64_t pb = (int64_t)c.r * S.pp_cap + cur;
+  rl_request(c, PEND_REALLOC, now, 1, S.pp_ing[pb], S.pp_size[pb],
+             S.pp_netlat[pb], S.pp_jid[pb], S.pp_dc[pb], 0);
+}
+
 // ---------------- the advance kernel ----------------
 template <int ALGO>
 __global__ void __launch_bounds__(THREADS_PER_BLOCK)
@@ -916,6 +1036,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   int64_t sbase = (int64_t)c.r * S.total_slots;
   long long n_events = 0;
   bool paused = false;
+  bool skip_loop = false;  // realloc chain pending: bypass the event loop
 
   // ---- carve the dynamic-LDS region ----
   // SUBW==64: per replica, Hot + s_finish mirror + x_time mirror.
@@ -1019,7 +1140,8 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           S.x_has_rl[at] = 1;
         }
       }
-    } else {  // PEND_DRAIN: one queued job, RL chose a target DC + g
+    } else if (pk == PEND_DRAIN) {
+      // one queued job, RL chose a target DC + g
       int src_d = S.pend_dc[c.r];
       int from_inf = S.pend_from_inf[c.r];
       int d_tgt = a_dc;
@@ -1033,15 +1155,55 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         rl_start_job(c, d_tgt, jt, size, netlat, jid, ing, n_sel, f, c.now,
                      s0, a_dc, a_g, mdc, mg, n_sel);
       }
+    } else {  // PEND_REALLOC: resume the pool entry at the cursor on its DC
+      int cur = S.pp_cursor[c.r];
+      int64_t pb = (int64_t)c.r * S.pp_cap + cur;
+      int d_src = S.pp_dc[pb];
+      if (c.free_gpus(d_src) <= 0) {
+        // no free GPUs: re-queue on the training queue (oracle's fix of
+        // reference Appendix A.6 job-stranding)
+        queue_push(c, d_src, 1, S.pp_size[pb], S.pp_netlat[pb],
+                   S.pp_jid[pb], S.pp_ing[pb]);
+      } else {
+        int n_rl = max(1, min(min(a_g + 1, c.free_gpus(d_src)), S.max_gpj));
+        double f = rl_energy_freq(c, d_src, 1, n_rl);
+        rl_start_job(c, d_src, 1, S.pp_size[pb], S.pp_netlat[pb],
+                     S.pp_jid[pb], S.pp_ing[pb], n_rl, f, c.now,
+                     &S.pp_s0[pb * S.obs_dim], S.pp_adc[pb], S.pp_ag[pb],
+                     mdc, mg, n_rl, S.pp_done[pb], S.pp_pcount[pb],
+                     S.pp_has_rl[pb]);
+      }
+      int nxt = cur + 1;
+      if (nxt < S.pp_count[c.r]) {
+        if (lane == 0) S.pp_cursor[c.r] = nxt;
+        store_fence();
+        // request the NEXT pool entry and skip straight to the hot-state
+        // writeback (n_events = max_ev empties the event loop; an early
+        // return here would LOSE the LDS-resident bookkeeping updates)
+        rl_request_realloc(c, c.now);
+        skip_loop = true;
+        paused = true;
+      } else {
+        if (lane == 0) {
+          S.pp_count[c.r] = 0;
+          S.pp_cursor[c.r] = 0;
+        }
+        store_fence();
+        // reallocation finished -> the deferred one-job queue drain
+        if (rl_try_drain_request(c, d_src, c.now)) {
+          skip_loop = true;
+          paused = true;
+        }
+      }
     }
-    if (lane == 0) {
+    if (!paused && lane == 0) {
       S.pend_kind[c.r] = PEND_NONE;
       S.req_flag[c.r] = REQ_IDLE;
     }
     store_fence();
   }
 
-  while (n_events < max_ev) {
+  while (!skip_loop && n_events < max_ev) {
     // ---- 1. next event: wave argmin over candidate sources ----
     // per-lane candidate: value + kind/idx
     double v = D_INF;
@@ -1316,6 +1478,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       float netlat = S.s_netlat[at];
       int jid = S.s_jid[at];
       int ingr = S.s_ing[at];
+      int pcount = S.s_pcount[at];
       double T = d_unit_time(n, fused, c.lc3(d, jt));
       if (lane == 0) {
         c.l_fin[slot] = D_INF;
@@ -1351,7 +1514,8 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         }
       }
       store_fence();
-      emit_job_row(c, d, jt, jid, ingr, size, fused, n, netlat, start, t_min);
+      emit_job_row(c, d, jt, jid, ingr, size, fused, n, netlat, start, t_min,
+                   pcount);
       rescan_dc_min(c, d);
       if (ALGO == A_CHSAC) {
         // record latency, then build the transition (reference :718-800)
@@ -1377,24 +1541,24 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           if (lane == 0) S.slot_has_rl[at] = 0;
           store_fence();
         }
-        // chsac drains AT MOST ONE queued job per finish, via a fresh policy
-        // action (reference :849-890)
-        double qsize;
-        float qnetlat;
-        int qjid, qing;
-        bool popped = false;
-        int from_inf = 0;
-        if (c.free_gpus(d) > 0) {
-          if (S.inf_priority && queue_pop(c, d, 0, qsize, qnetlat, qjid, qing)) {
-            popped = true;
-            from_inf = 1;
-          } else if (queue_pop(c, d, 1, qsize, qnetlat, qjid, qing)) {
-            popped = true;
+        // elastic scaling: on a training completion with other training jobs
+        // still running in this DC, preempt them all and reallocate via
+        // fresh RL actions (reference :829-837; gated to chsac + flag)
+        if (S.elastic && jt == 1) {
+          int n_train = 0;
+          for (int k = S.slot_off[d]; k < S.slot_off[d + 1]; ++k)
+            if (S.s_gpus[sbase + k] != 0 && S.s_jtype[sbase + k] == 1)
+              ++n_train;
+          if (n_train > 1) {
+            int cnt = rl_elastic_preempt_all(c, d, t_min);
+            if (cnt > 0) {
+              rl_request_realloc(c, t_min);
+              paused = true;
+              break;
+            }
           }
         }
-        if (popped) {
-          rl_request(c, PEND_DRAIN, t_min, from_inf ? 0 : 1, qing, qsize,
-                     qnetlat, qjid, d, from_inf);
+        if (rl_try_drain_request(c, d, t_min)) {
           paused = true;
           break;
         }
@@ -1538,6 +1702,8 @@ class BatchedSimHip {
     S_.s_gpus = reinterpret_cast<short*>(t_["s_gpus"].data_ptr<int16_t>());
     S_.s_jtype = reinterpret_cast<char*>(t_["s_jtype"].data_ptr<int8_t>());
     S_.s_ing = reinterpret_cast<char*>(t_["s_ing"].data_ptr<int8_t>());
+    T_PTR(s_done, double);
+    S_.s_pcount = reinterpret_cast<unsigned char*>(t_["s_pcount"].data_ptr<uint8_t>());
     T_PTR(x_time, double); T_PTR(x_size, double); T_PTR(x_netlat, float);
     T_PTR(x_jid, int);
     S_.x_dc = reinterpret_cast<char*>(t_["x_dc"].data_ptr<int8_t>());
@@ -1570,6 +1736,8 @@ class BatchedSimHip {
     S_.obs_dim = cfg.contains("obs_dim") ? cfg["obs_dim"].cast<int>() : 0;
     S_.sla_p99_ms = cfg.contains("sla_p99_ms") ? cfg["sla_p99_ms"].cast<double>() : 500.0;
     S_.tr_cap = cfg.contains("tr_cap") ? cfg["tr_cap"].cast<int>() : 0;
+    S_.elastic = cfg.contains("elastic") ? cfg["elastic"].cast<int>() : 0;
+    S_.pp_cap = cfg.contains("pp_cap") ? cfg["pp_cap"].cast<int>() : 0;
     if (S_.algo == A_CHSAC) {
       T_PTR(req_flag, int); T_PTR(req_obs, float); T_PTR(req_mdc, int);
       T_PTR(req_mg, int); T_PTR(resp_dc, int); T_PTR(resp_g, int);
@@ -1583,6 +1751,17 @@ class BatchedSimHip {
       S_.slot_mg = reinterpret_cast<unsigned char*>(t_["slot_mg"].data_ptr<uint8_t>());
       S_.slot_has_rl = reinterpret_cast<unsigned char*>(t_["slot_has_rl"].data_ptr<uint8_t>());
       S_.slot_nrew = reinterpret_cast<unsigned char*>(t_["slot_nrew"].data_ptr<uint8_t>());
+      T_PTR(pp_count, int); T_PTR(pp_cursor, int);
+      T_PTR(pp_size, double); T_PTR(pp_done, double); T_PTR(pp_netlat, float);
+      T_PTR(pp_jid, int);
+      S_.pp_ing = reinterpret_cast<unsigned char*>(t_["pp_ing"].data_ptr<uint8_t>());
+      S_.pp_dc = reinterpret_cast<unsigned char*>(t_["pp_dc"].data_ptr<uint8_t>());
+      S_.pp_pcount = reinterpret_cast<unsigned char*>(t_["pp_pcount"].data_ptr<uint8_t>());
+      T_PTR(pp_s0, float);
+      S_.pp_adc = reinterpret_cast<unsigned char*>(t_["pp_adc"].data_ptr<uint8_t>());
+      S_.pp_ag = reinterpret_cast<unsigned char*>(t_["pp_ag"].data_ptr<uint8_t>());
+      S_.pp_nrew = reinterpret_cast<unsigned char*>(t_["pp_nrew"].data_ptr<uint8_t>());
+      S_.pp_has_rl = reinterpret_cast<unsigned char*>(t_["pp_has_rl"].data_ptr<uint8_t>());
       T_PTR(x_s0, float);
       S_.x_nsel = reinterpret_cast<short*>(t_["x_nsel"].data_ptr<int16_t>());
       S_.x_adc = reinterpret_cast<unsigned char*>(t_["x_adc"].data_ptr<uint8_t>());

@@ -457,3 +457,35 @@ def test_subwave_variants_bit_identical():
         assert torch.equal(e64.t[key].cpu(), e8.t[key].cpu()), key
     assert torch.equal(e64.t["energy_j"].cpu(), e8.t["energy_j"].cpu())
     assert torch.equal(e64.t["sum_lat"].cpu(), e8.t["sum_lat"].cpu())
+
+
+@needs_gpu
+def test_chsac_elastic_scaling_on_gpu(tmp_path):
+    """Elastic scaling on the batched engine: training completions with other
+    training jobs running trigger preempt-all + RL reallocation; preempt
+    counts surface in the job log and invariants hold (GPU capability parity
+    with the oracle's elastic path)."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="off", rate=0.0)
+    trn = ArrivalProcess(mode="poisson", rate=0.5)
+    out = str(tmp_path / "el")
+    eng = BatchedEngine(sc, inf, trn, algo="chsac_af", replicas=16,
+                        duration=1500.0, log_interval=10.0, out_dir=out,
+                        seed=5, enable_logs=True, elastic_scaling=True,
+                        rl_warmup=10**9,  # act-only, no training needed here
+                        events_per_launch=20000)
+    st = eng.run()
+    assert int(eng.t["err"].max().item()) == 0
+    assert st["jobs_completed"] > 0
+    assert eng.validate_state()
+    with open(os.path.join(out, "job_log.csv")) as fh:
+        rows = list(csv.DictReader(fh))
+    assert rows
+    total_preempts = sum(int(r["preempt_count"]) for r in rows)
+    assert total_preempts > 0, "elastic scaling never preempted"
+    # preempted jobs keep their identity and complete exactly once
+    jids = [r["jid"] for r in rows]
+    assert len(jids) == len(set(jids))
