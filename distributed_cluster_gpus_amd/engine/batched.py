@@ -262,8 +262,9 @@ class BatchedEngine:
             t["x_mdc"] = torch.zeros((R, tcap), **u8)
             t["x_mg"] = torch.zeros((R, tcap), **u8)
             t["x_has_rl"] = torch.zeros((R, tcap), **u8)
-            # elastic-scaling preempted-job pool
-            pp_cap = 64
+            # elastic-scaling preempted-job pool: a DC can run up to
+            # total_gpus concurrent 1-GPU training jobs, all preemptible
+            pp_cap = int(scenario.total_gpus.max())
             self._pp_cap = pp_cap
             u8_ = dict(dtype=torch.uint8, device=dev)
             t["pp_count"] = torch.zeros(R, **i32)
