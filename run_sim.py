@@ -161,6 +161,17 @@ def main(argv=None):
         eng.rl.load(args.rl_resume)
 
     stats = eng.run()
+    if args.engine == "batched" and args.replicas > 1:
+        # Monte-Carlo population report: distributions over the replica
+        # ensemble (a capability the scalar reference cannot express)
+        from distributed_cluster_gpus_amd.analysis.montecarlo import \
+            population_report
+        rep = population_report(eng, out_dir=os.path.join(out_dir, "population"))
+        e = rep.get("total_energy_kJ")
+        if e:
+            print(f"[population] {rep['replicas']} replicas: total energy "
+                  f"{e['mean']:.1f} kJ ± {e['stderr']:.2f} (95% CI "
+                  f"[{e['ci_lo']:.1f}, {e['ci_hi']:.1f}])")
     if args.rl_checkpoint and getattr(eng, "rl", None) is not None:
         eng.rl.save(args.rl_checkpoint)
         print(f"RL checkpoint saved to {args.rl_checkpoint}")

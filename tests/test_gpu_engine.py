@@ -489,3 +489,21 @@ def test_chsac_elastic_scaling_on_gpu(tmp_path):
     # preempted jobs keep their identity and complete exactly once
     jids = [r["jid"] for r in rows]
     assert len(jids) == len(set(jids))
+
+
+@needs_gpu
+def test_population_report(tmp_path):
+    """Monte-Carlo population statistics over the replica ensemble."""
+    from distributed_cluster_gpus_amd.analysis.montecarlo import (
+        population_frame, population_report)
+    eng = make_engine(replicas=128, duration=90.0)
+    eng.run()
+    df = population_frame(eng)
+    assert len(df) == 128
+    rep = population_report(eng, out_dir=str(tmp_path / "pop"))
+    e = rep["total_energy_kJ"]
+    assert e["std"] > 0 and e["ci_lo"] < e["mean"] < e["ci_hi"]
+    assert e["p01"] <= e["p50"] <= e["p99"]
+    assert os.path.exists(os.path.join(str(tmp_path / "pop"), "population.csv"))
+    assert os.path.exists(os.path.join(str(tmp_path / "pop"),
+                                       "population_stats.csv"))
