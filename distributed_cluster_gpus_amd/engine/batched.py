@@ -318,11 +318,6 @@ class BatchedEngine:
                                      n_dc=n_dc,
                                      n_g=int(scenario.policy.max_gpus_per_job),
                                      device=str(dev), seed=seed)
-            # bit-expansion LUT for mask bytes -> bool vectors
-            ar = torch.arange(max(n_dc, int(scenario.policy.max_gpus_per_job)),
-                              device=dev)
-            self._bitpos = ar
-
         self.t = t
         self.arrival_inf, self.arrival_trn = arrival_inf, arrival_trn
 
