@@ -507,3 +507,28 @@ def test_population_report(tmp_path):
     assert os.path.exists(os.path.join(str(tmp_path / "pop"), "population.csv"))
     assert os.path.exists(os.path.join(str(tmp_path / "pop"),
                                        "population_stats.csv"))
+
+
+@needs_gpu
+def test_train_rl_dp_driver_single_gpu(tmp_path):
+    """scripts/train_rl_dp.py end-to-end on one GPU (the 8-GPU DP path minus
+    the collectives, which the gloo tests pin)."""
+    ckpt = str(tmp_path / "dp.pt")
+    r = subprocess.run([sys.executable,
+                        os.path.join(REPO, "scripts", "train_rl_dp.py"),
+                        "--replicas-per-gpu", "48", "--duration", "150",
+                        "--warmup", "100", "--batch", "64",
+                        "--train-interval", "64", "--checkpoint", ckpt,
+                        "--replay-npz", str(tmp_path / "ds.npz")],
+                       capture_output=True, text=True, timeout=900, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["jobs_completed"] > 0 and out["rl_updates"] > 0
+    assert os.path.exists(ckpt)
+    # offline dataset round-trips into the offline trainer
+    from distributed_cluster_gpus_amd.rl.offline import train_offline
+    agent, stats = train_offline(str(tmp_path / "ds.npz"), epochs=1,
+                                 batch_size=64, device="cuda",
+                                 constraints={"latency_p99": 500.0})
+    assert stats and np.isfinite(stats[-1]["loss_critic"])
