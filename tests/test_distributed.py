@@ -96,6 +96,59 @@ def _worker_ddp_sac(rank, world, port, q):
         q.put((rank, str(e)))
 
 
+def _worker_dp_sync(rank, world, port, q):
+    """Ranks feed dp_sync_step deliberately UNEQUAL local state (different
+    replay sizes, transition counts, done flags) and must still agree on the
+    derived (err, all_done, min_replay, total_new) tuple — the control word
+    that keeps per-launch SAC train-step counts identical across ranks."""
+    _init(rank, world, port)
+    try:
+        import torch.distributed as dist
+        from distributed_cluster_gpus_amd.parallel.dist import dp_sync_step
+        # launch 1: rank 0 is done, rank 1 is not; different counters
+        out1 = dp_sync_step(err=0, local_done=(rank == 0),
+                            replay_size=100 + 50 * rank, n_new=10 * (rank + 1))
+        # launch 2: rank 1 reports an error flag
+        out2 = dp_sync_step(err=4 if rank == 1 else 0, local_done=True,
+                            replay_size=1000, n_new=0)
+        q.put((rank, out1, out2))
+        dist.barrier()
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, str(e), None))
+
+
+def _worker_dp_train_cadence(rank, world, port, q):
+    """Both ranks must execute the SAME number of train_step collectives when
+    their local transition streams differ (the advisor's round-1 mispairing
+    scenario), derived from globally reduced counters."""
+    _init(rank, world, port)
+    try:
+        import torch.distributed as dist
+        from distributed_cluster_gpus_amd.parallel.dist import dp_sync_step
+        interval, warmup = 4, 8
+        backlog, steps_run = 0, []
+        # rank 0 produces transitions twice as fast; rank 1's replay crosses
+        # warmup later
+        local_replay = 0
+        for launch in range(6):
+            n_new = (2 if rank == 0 else 1) * 3
+            local_replay += n_new
+            _, _, min_replay, tr_total = dp_sync_step(
+                0, launch == 5, local_replay, n_new)
+            backlog += tr_total
+            steps = 0
+            if min_replay >= warmup:
+                steps = min(64, backlog // (interval * world))
+                backlog -= steps * interval * world
+            steps_run.append(steps)
+        q.put((rank, steps_run))
+        dist.barrier()
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, str(e)))
+
+
 def _spawn(fn, world=2, port=29801):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
@@ -128,6 +181,28 @@ def test_ddp_sac_parameters_stay_synced():
         ws[rank] = w
     # critic params must be bitwise identical after synced-grad training
     np.testing.assert_array_equal(ws[0], ws[1])
+
+
+def test_dp_sync_step_agrees_across_ranks():
+    res = _spawn(_worker_dp_sync, port=29831)
+    outs = {}
+    for rank, out1, out2 in res:
+        assert not isinstance(out1, str), f"rank {rank} failed: {out1}"
+        outs[rank] = (out1, out2)
+    assert outs[0] == outs[1]
+    out1, out2 = outs[0]
+    assert out1 == (0, False, 100, 30)   # min replay, summed transitions
+    assert out2[0] == 4 and out2[1] is True  # error surfaced everywhere
+
+
+def test_dp_train_cadence_identical_across_ranks():
+    res = _spawn(_worker_dp_train_cadence, port=29841)
+    steps = {}
+    for rank, s in res:
+        assert not isinstance(s, str), f"rank {rank} failed: {s}"
+        steps[rank] = s
+    assert steps[0] == steps[1], "per-launch SAC step counts diverged"
+    assert sum(steps[0]) > 0
 
 
 def test_replica_shard_partition():
