@@ -152,6 +152,9 @@ class BatchedEngine:
         t["dc_min_slot"] = torch.full((R, n_dc), -1, **i32)
         t["s_finish"] = torch.full((R, total_slots), INF, **f64)
         t["s_start"] = torch.zeros((R, total_slots), **f64)
+        t["s_lastupd"] = torch.zeros((R, total_slots), **f64)
+        t["s_seq"] = torch.zeros((R, total_slots), **i32)
+        t["seq_ctr"] = torch.zeros(R, **i32)
         t["s_size"] = torch.zeros((R, total_slots), **f64)
         t["s_fused"] = torch.zeros((R, total_slots), **f64)
         t["s_netlat"] = torch.zeros((R, total_slots), **f32)
@@ -172,6 +175,10 @@ class BatchedEngine:
         t["q_head"] = torch.zeros((R, n_dc, 2), **i32)
         t["q_len"] = torch.zeros((R, n_dc, 2), **i32)
         t["q_size"] = torch.zeros((R, n_dc, 2, qcap), **f64)
+        t["q_enq"] = torch.zeros((R, n_dc, 2, qcap), **f64)
+        # cap_greedy pass-snapshot scratch (frozen task frequencies)
+        t["snap_f"] = torch.zeros(
+            (R, total_slots) if algo == "cap_greedy" else (1, 1), **f64)
         # queue aux fields (net latency / jid / ingress) are only consumed by
         # the logging replica's job rows -> single-replica allocation
         t["q_netlat"] = torch.zeros((n_dc, 2, qcap), **f32)
@@ -201,7 +208,7 @@ class BatchedEngine:
             if self.log_replica >= 0 else 1
         self._jl_chunks = []  # host-side drained job-row chunks (np arrays)
         t["cl_count"] = torch.zeros(1, **i32)
-        t["cl_rows"] = torch.zeros((cl_cap, 16), **f64)
+        t["cl_rows"] = torch.zeros((cl_cap, 15), **f64)
         t["jl_count"] = torch.zeros(1, **i32)
         t["jl_rows"] = torch.zeros((jl_cap, 11), **f64)
 
@@ -260,6 +267,7 @@ class BatchedEngine:
             t["pend_jt"] = torch.zeros(R, **i32)
             t["pend_dc"] = torch.zeros(R, **i32)
             t["pend_from_inf"] = torch.zeros(R, **i32)
+            t["pend_enq"] = torch.zeros(R, **f64)
             u8 = dict(dtype=torch.uint8, device=dev)
             t["slot_s0"] = torch.zeros((R, total_slots, obs_dim), **f32)
             t["slot_adc"] = torch.zeros((R, total_slots), **u8)

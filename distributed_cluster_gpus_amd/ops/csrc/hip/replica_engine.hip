@@ -117,6 +117,11 @@ struct EngineDesc {
   char* s_jtype;
   char* s_ing;
   double* s_done;                 // units completed before this (re)start
+  double* s_lastupd;              // last progress-update time (reference
+                                  //   Job.last_update; advanced by the cap
+                                  //   controller's reschedule)
+  int* s_seq;                     // monotone start sequence — running_jobs
+                                  //   dict insertion order (atom tie-break)
   unsigned char* s_pcount;        // preemptions experienced
   // in-flight transfers [r][tcap]
   double* x_time;                 // INF = empty
@@ -131,6 +136,8 @@ struct EngineDesc {
   int* q_head;
   int* q_len;
   double* q_size;                 // [r][dc][2][qcap] (f64 for exact parity)
+  double* q_enq;                  // [r][dc][2][qcap] enqueue time (queueing-
+                                  //   delay metric; FIFO-paired with pops)
   float* q_netlat;                // [dc][2][qcap] (log replica only)
   int* q_jid;                     // [dc][2][qcap] (log replica only)
   char* q_ing;                    // [dc][2][qcap] (log replica only)
@@ -144,14 +151,20 @@ struct EngineDesc {
   long long* jobs_done_inf;
   double* sum_lat;
   double* sum_lat_inf;
-  double* sum_wait;               // queueing delay (start - dc arrival)
-  // logging (one designated replica)
+  double* sum_wait;               // summed queueing delay (job start minus
+                                  //   its enqueue at the DC; 0 for jobs that
+                                  //   start straight off the WAN transfer)
+  int* seq_ctr;                   // [r] job-start sequence counter
+  double* snap_f;                 // [r][total_slots] cap_greedy pass snapshot
+                                  //   (allocated only for algo==cap_greedy)
+  // logging (one designated replica); the host drains both buffers
+  // chunk-wise between launches, so caps bound one launch's production only
   int log_replica;                // -1 = off
   int cl_cap, jl_cap;
   int* cl_count;                  // [1]
-  double* cl_rows;                // [cl_cap][16]
+  double* cl_rows;                // [cl_cap][15]
   int* jl_count;                  // [1]
-  double* jl_rows;                // [jl_cap][10]
+  double* jl_rows;                // [jl_cap][11]
   // ===== arrival-trace replay mode (single-replica exact-parity testing:
   // arrivals come from a host-recorded (time, size) FIFO per stream instead
   // of the Philox draws) =====
@@ -181,6 +194,7 @@ struct EngineDesc {
   int* pend_jt;                   // [r]
   int* pend_dc;                   // [r] (drain: source DC)
   int* pend_from_inf;             // [r] (drain: which queue the job came from)
+  double* pend_enq;               // [r] (drain: the popped job's enqueue time)
   // per-job RL traces (state0 / action / masks at selection time)
   float* slot_s0;                 // [r][total_slots][obs_dim]
   unsigned char* slot_adc;        // [r][total_slots]
@@ -349,6 +363,7 @@ __device__ void start_job(Ctx& c, int d, int jt, double size, float netlat,
   if (c.lane == 0) {
     c.l_fin[cand] = finish;
     S.s_start[base + cand] = now;
+    S.s_lastupd[base + cand] = now;
     S.s_size[base + cand] = size;
     S.s_fused[base + cand] = f;
     S.s_netlat[base + cand] = netlat;
@@ -358,6 +373,9 @@ __device__ void start_job(Ctx& c, int d, int jt, double size, float netlat,
     S.s_ing[base + cand] = (char)ing;
     S.s_done[base + cand] = 0.0;
     S.s_pcount[base + cand] = 0;
+    int sq = S.seq_ctr[c.r] + 1;
+    S.seq_ctr[c.r] = sq;
+    S.s_seq[base + cand] = sq;
     c.hs->busy[d] += n;
     c.hs->n_running[d] += 1;
     c.hs->p_active[d] += d_job_power(n, f, c.pc3(d, jt));
@@ -443,9 +461,10 @@ __device__ void decide_nf(Ctx& c, int d, int jt, float size, double now,
   }
 }
 
-// queue ops (wave-uniform; lane-0 writes)
+// queue ops (wave-uniform; lane-0 writes).  `enq` is the job's enqueue time
+// at this DC — carried per entry so pops can credit the queueing delay.
 __device__ bool queue_push(Ctx& c, int d, int jt, double size, float netlat,
-                           int jid, int ing) {
+                           int jid, int ing, double enq) {
   const EngineDesc& S = *c.S;
   int ql = d * 2 + jt;
   int len = c.hs->q_len[ql];
@@ -457,6 +476,7 @@ __device__ bool queue_push(Ctx& c, int d, int jt, double size, float netlat,
   if (c.lane == 0) {
     int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
     S.q_size[at] = size;
+    S.q_enq[at] = enq;
     if (c.r == S.log_replica) {
       int64_t aux = ((int64_t)ql) * S.qcap + pos;
       S.q_netlat[aux] = netlat;
@@ -470,7 +490,7 @@ __device__ bool queue_push(Ctx& c, int d, int jt, double size, float netlat,
 }
 
 __device__ bool queue_push_front(Ctx& c, int d, int jt, double size,
-                                 float netlat, int jid, int ing) {
+                                 float netlat, int jid, int ing, double enq) {
   const EngineDesc& S = *c.S;
   int ql = d * 2 + jt;
   int len = c.hs->q_len[ql];
@@ -482,6 +502,7 @@ __device__ bool queue_push_front(Ctx& c, int d, int jt, double size,
   if (c.lane == 0) {
     int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
     S.q_size[at] = size;
+    S.q_enq[at] = enq;
     if (c.r == S.log_replica) {
       int64_t aux = ((int64_t)ql) * S.qcap + pos;
       S.q_netlat[aux] = netlat;
@@ -496,7 +517,7 @@ __device__ bool queue_push_front(Ctx& c, int d, int jt, double size,
 }
 
 __device__ bool queue_pop(Ctx& c, int d, int jt, double& size, float& netlat,
-                          int& jid, int& ing) {
+                          int& jid, int& ing, double& enq) {
   const EngineDesc& S = *c.S;
   int ql = d * 2 + jt;
   int len = c.hs->q_len[ql];
@@ -504,6 +525,7 @@ __device__ bool queue_pop(Ctx& c, int d, int jt, double& size, float& netlat,
   int pos = c.hs->q_head[ql];
   int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
   size = S.q_size[at];
+  enq = S.q_enq[at];
   if (c.r == S.log_replica) {
     int64_t aux = ((int64_t)ql) * S.qcap + pos;
     netlat = S.q_netlat[aux];
@@ -525,12 +547,12 @@ template <int ALGO>
 __device__ void drain_queues(Ctx& c, int d, double now) {
   const EngineDesc& S = *c.S;
   while (c.free_gpus(d) > 0) {
-    double size;
+    double size, enq;
     float netlat;
     int jid, ing;
     int jt;
-    if (S.inf_priority && queue_pop(c, d, 0, size, netlat, jid, ing)) jt = 0;
-    else if (queue_pop(c, d, 1, size, netlat, jid, ing)) jt = 1;
+    if (S.inf_priority && queue_pop(c, d, 0, size, netlat, jid, ing, enq)) jt = 0;
+    else if (queue_pop(c, d, 1, size, netlat, jid, ing, enq)) jt = 1;
     else break;
     int n; double f;
     if (ALGO == A_CARBON_COST) {
@@ -551,6 +573,7 @@ __device__ void drain_queues(Ctx& c, int d, double now) {
     }
     n = max(1, min(n, c.free_gpus(d)));
     start_job(c, d, jt, size, netlat, jid, ing, n, f, now);
+    if (c.lane == 0) S.sum_wait[c.r] += fmax(0.0, now - enq);
   }
 }
 
@@ -585,7 +608,7 @@ __device__ void emit_cluster_rows(Ctx& c, double now) {
     if (c.lane == 0) {
       int idx = *S.cl_count;
       if (idx < S.cl_cap) {
-        double* row = &S.cl_rows[(int64_t)idx * 16];
+        double* row = &S.cl_rows[(int64_t)idx * 15];
         double begin = c.hs->util_begin[d];
         double elapsed = fmax(1e-9, now - (begin >= 0 ? begin : now));
         row[0] = now;
@@ -604,7 +627,6 @@ __device__ void emit_cluster_rows(Ctx& c, double now) {
         row[12] = c.hs->acc_unit[d];
         row[13] = c.dc_power(d);
         row[14] = c.hs->energy[d];
-        row[15] = 0.0;
         *S.cl_count = idx + 1;
       } else {
         atomicOr(&S.err[c.r], ERR_LOG_OVF);
@@ -634,73 +656,143 @@ __device__ void emit_job_row(Ctx& c, int d, int jt, int jid, int ing,
   store_fence();
 }
 
-// cap_greedy controller step at log ticks (reference :248-315, v1: repeatedly
-// apply the currently-cheapest single down-step atom with exact re-estimation;
-// distributionally equivalent to the reference's per-pass sorted ladders)
+// cap_greedy controller step at log ticks — reference-exact sorted-snapshot
+// semantics (freq_load_agg.py:44-80 + simulator_paper_multi.py:229-315):
+//   outer pass: snapshot every running job's f; build the full DOWN-ladder
+//   atom list from that snapshot (atom (slot, k) = step lv[k] -> lv[k-1] for
+//   k <= i0, rho frozen at snapshot ladder values);
+//   inner loop: apply atoms in ascending (rho, task-order, step-order) —
+//   emulated as a selection walk so no materialized sort is needed; an atom
+//   whose target f is not strictly below the job's LIVE f is skipped (a
+//   cheaper deeper atom may already have jumped the job lower); each applied
+//   atom reschedules the job with the reference's exact progress arithmetic
+//   and re-estimates total power before the next deficit check.
+// Task order = DC-major, then running_jobs insertion order (s_seq); step
+// order = i0-k ascending — exactly the reference's stable rho sort.
 __device__ void cap_greedy_control(Ctx& c, double now) {
   const EngineDesc& S = *c.S;
-  double f_min = S.freq_levels[0];
-  for (int k = 1; k < S.n_freq; ++k) f_min = fmin(f_min, S.freq_levels[k]);
+  // sorted ladder (atoms_for_task sorts freq_levels)
+  double lv[MAX_FREQ];
+  for (int i = 0; i < S.n_freq; ++i) lv[i] = S.freq_levels[i];
+  for (int i = 1; i < S.n_freq; ++i) {
+    double x = lv[i];
+    int j = i - 1;
+    while (j >= 0 && lv[j] > x) { lv[j + 1] = lv[j]; --j; }
+    lv[j + 1] = x;
+  }
+  double f_min = lv[0];
+  double totalP = 0;
+  for (int d = 0; d < S.n_dc; ++d) totalP += c.dc_power(d);
+  if (totalP <= S.power_cap - 5.0) return;  // hysteresis (reference :235-237)
+  double deficit = fmax(0.0, totalP - S.power_cap);
+  if (deficit <= 1e-6) return;
+  int64_t base = (int64_t)c.r * S.total_slots;
   int guard = 10000;
-  while (guard-- > 0) {
-    double totalP = 0;
-    for (int d = 0; d < S.n_dc; ++d) totalP += c.dc_power(d);
-    double deficit = totalP - S.power_cap;
-    if (deficit <= 1e-6) break;
-    // find min-rho single-step-down atom over all running jobs (lane-strided)
-    int64_t base = (int64_t)c.r * S.total_slots;
-    double best_rho = D_INF;
-    int best_slot = -1;
-    double best_fto = 0;
+  while (deficit > 1e-6 && guard-- > 0) {
+    // ---- snapshot: freeze each cappable task's f for this pass ----
+    int have = 0;
     for (int k = c.lane; k < S.total_slots; k += SUBW) {
-      if (c.l_fin[k] >= D_INF) continue;
-      int d = S.slot_dc[k];
-      double fu = S.s_fused[base + k];
-      if (fu <= f_min + 1e-12) continue;
-      // nearest ladder index
-      int i0 = 0;
-      double bd = 1e300;
-      for (int q = 0; q < S.n_freq; ++q) {
-        double diff = fabs(S.freq_levels[q] - fu);
-        if (diff < bd) { bd = diff; i0 = q; }
+      double sf = 0.0;
+      if (c.l_fin[k] < D_INF) {
+        double fu = S.s_fused[base + k];
+        if (fu > f_min + 1e-12) { sf = fu; have = 1; }
       }
-      if (i0 == 0) continue;
-      int jt = S.s_jtype[base + k];
-      int n = S.s_gpus[base + k];
-      double f_to = S.freq_levels[i0 - 1];
-      double v1 = 1.0 / d_unit_time(n, S.freq_levels[i0], c.lc3(d, jt));
-      double p1 = d_job_power(n, S.freq_levels[i0], c.pc3(d, jt));
-      double v2 = 1.0 / d_unit_time(n, f_to, c.lc3(d, jt));
-      double p2 = d_job_power(n, f_to, c.pc3(d, jt));
-      double dV = fmax(0.0, v1 - v2), dP = fmax(0.0, p1 - p2);
-      if (dV <= 0 || dP < 0) continue;
-      double rho = dP / dV;
-      if (rho < best_rho) { best_rho = rho; best_slot = k; best_fto = f_to; }
-    }
-    int wl;
-    double rho = wave_argmin_f64(best_rho, wl);
-    if (rho >= D_INF) break;
-    best_slot = __shfl(best_slot, wl, 64);
-    best_fto = __shfl(best_fto, wl, 64);
-    // apply: advance progress implicitly by recomputing remaining time
-    int d = S.slot_dc[best_slot];
-    int jt = S.s_jtype[base + best_slot];
-    int n = S.s_gpus[base + best_slot];
-    double old_f = S.s_fused[base + best_slot];
-    double T_old = d_unit_time(n, old_f, c.lc3(d, jt));
-    double T_new = d_unit_time(n, best_fto, c.lc3(d, jt));
-    double finish_old = c.l_fin[best_slot];
-    double remaining_units = fmax(0.0, (finish_old - now)) / fmax(T_old, 1e-12);
-    double finish_new = now + remaining_units * T_new;
-    if (c.lane == 0) {
-      c.hs->p_active[d] += d_job_power(n, best_fto, c.pc3(d, jt)) -
-                           d_job_power(n, old_f, c.pc3(d, jt));
-      c.hs->sum_tpt[d] += 1.0 / T_new - 1.0 / T_old;
-      S.s_fused[base + best_slot] = best_fto;
-      c.l_fin[best_slot] = finish_new;
+      S.snap_f[base + k] = sf;
     }
     store_fence();
-    rescan_dc_min(c, d);
+    if (wave_sum_i32(have) == 0) break;  // no tasks
+    bool applied_any = false;
+    bool exhausted = false;
+    // frozen-order selection walk: strictly-after (last_rho, last_key)
+    double last_rho = -D_INF;
+    long long last_key = -1;
+    while (deficit > 1e-6) {
+      double best_rho = D_INF;
+      long long best_key = LLONG_MAX;
+      int best_slot = -1, best_k = -1;
+      for (int k = c.lane; k < S.total_slots; k += SUBW) {
+        double sf = S.snap_f[base + k];
+        if (sf <= 0.0) continue;
+        int i0 = 0;
+        double bd = 1e300;
+        for (int q = 0; q < S.n_freq; ++q) {
+          double df = fabs(lv[q] - sf);
+          if (df < bd) { bd = df; i0 = q; }
+        }
+        if (i0 == 0) continue;
+        int d = S.slot_dc[k];
+        int jt = S.s_jtype[base + k];
+        int n = S.s_gpus[base + k];
+        long long seq = S.s_seq[base + k];
+        double curV = 1.0 / d_unit_time(n, lv[i0], c.lc3(d, jt));
+        double curP = d_job_power(n, lv[i0], c.pc3(d, jt));
+        for (int kk = i0; kk >= 1; --kk) {
+          double V2 = 1.0 / d_unit_time(n, lv[kk - 1], c.lc3(d, jt));
+          double P2 = d_job_power(n, lv[kk - 1], c.pc3(d, jt));
+          double dV = fmax(0.0, curV - V2), dP = fmax(0.0, curP - P2);
+          if (dV > 0 && dP >= 0) {
+            double rho = dP / dV;
+            long long key = ((long long)d << 40) | (seq << 8) | (i0 - kk);
+            bool after = rho > last_rho || (rho == last_rho && key > last_key);
+            if (after && (rho < best_rho ||
+                          (rho == best_rho && key < best_key))) {
+              best_rho = rho; best_key = key; best_slot = k; best_k = kk;
+            }
+          }
+          curV = V2;
+          curP = P2;
+        }
+      }
+      // subgroup reduction on (rho, key)
+#pragma unroll
+      for (int off = SUBW / 2; off > 0; off >>= 1) {
+        double orho = __shfl_xor(best_rho, off, 64);
+        long long okey = __shfl_xor(best_key, off, 64);
+        int oslot = __shfl_xor(best_slot, off, 64);
+        int ok = __shfl_xor(best_k, off, 64);
+        if (orho < best_rho || (orho == best_rho && okey < best_key)) {
+          best_rho = orho; best_key = okey; best_slot = oslot; best_k = ok;
+        }
+      }
+      if (best_rho >= D_INF) { exhausted = true; break; }  // atom list done
+      last_rho = best_rho;
+      last_key = best_key;
+      // skip unless the atom still goes strictly DOWN from the live f
+      double cur_f = S.s_fused[base + best_slot];
+      double f_to = lv[best_k - 1];
+      if (f_to >= cur_f - 1e-12) continue;
+      // ---- apply: the reference's exact reschedule arithmetic ----
+      int d = S.slot_dc[best_slot];
+      int jt = S.s_jtype[base + best_slot];
+      int n = S.s_gpus[base + best_slot];
+      double T_old = d_unit_time(n, cur_f, c.lc3(d, jt));
+      double rate_old = 1.0 / fmax(T_old, 1e-9);
+      double size = S.s_size[base + best_slot];
+      double dt = fmax(0.0, now - S.s_lastupd[base + best_slot]);
+      double done = fmin(size, S.s_done[base + best_slot] + rate_old * dt);
+      double units_left = fmax(0.0, size - done);
+      double T_new = d_unit_time(n, f_to, c.lc3(d, jt));
+      double rate_new = 1.0 / fmax(T_new, 1e-9);
+      double finish_new = now + units_left / fmax(rate_new, 1e-9);
+      if (c.lane == 0) {
+        S.s_done[base + best_slot] = done;
+        S.s_lastupd[base + best_slot] = now;
+        S.s_fused[base + best_slot] = f_to;
+        c.l_fin[best_slot] = finish_new;
+        c.hs->p_active[d] += d_job_power(n, f_to, c.pc3(d, jt)) -
+                             d_job_power(n, cur_f, c.pc3(d, jt));
+        c.hs->sum_tpt[d] += rate_new - rate_old;
+      }
+      store_fence();
+      rescan_dc_min(c, d);
+      applied_any = true;
+      // exact total-power re-estimate after each applied atom (:300-307)
+      totalP = 0;
+      for (int dd = 0; dd < S.n_dc; ++dd) totalP += c.dc_power(dd);
+      deficit = fmax(0.0, totalP - S.power_cap);
+    }
+    if (!applied_any) break;
+    if (exhausted && deficit > 1e-6) continue;  // rebuild snapshot, next pass
   }
 }
 
@@ -784,7 +876,7 @@ __device__ void rl_build_masks(Ctx& c, int& mdc, int& mg) {
 // write an action request and stash the paused-event context
 __device__ void rl_request(Ctx& c, int kind, double now, int jt, int ing,
                            double size, float netlat, int jid, int src_dc,
-                           int from_inf) {
+                           int from_inf, double enq) {
   const EngineDesc& S = *c.S;
   rl_build_obs(c, now, &S.req_obs[(int64_t)c.r * S.obs_dim]);
   int mdc, mg;
@@ -800,6 +892,7 @@ __device__ void rl_request(Ctx& c, int kind, double now, int jt, int ing,
     S.pend_jt[c.r] = jt;
     S.pend_dc[c.r] = src_dc;
     S.pend_from_inf[c.r] = from_inf;
+    S.pend_enq[c.r] = enq;
     S.req_flag[c.r] = REQ_PENDING;
   }
   store_fence();
@@ -855,6 +948,7 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
   if (c.lane == 0) {
     c.l_fin[cand] = finish;
     S.s_start[base + cand] = now;
+    S.s_lastupd[base + cand] = now;
     S.s_size[base + cand] = size;
     S.s_fused[base + cand] = f;
     S.s_netlat[base + cand] = netlat;
@@ -864,6 +958,9 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
     S.s_ing[base + cand] = (char)ing;
     S.s_done[base + cand] = units_done;
     S.s_pcount[base + cand] = (unsigned char)pcount;
+    int sq = S.seq_ctr[c.r] + 1;
+    S.seq_ctr[c.r] = sq;
+    S.s_seq[base + cand] = sq;
     S.slot_adc[base + cand] = (unsigned char)a_dc;
     S.slot_ag[base + cand] = (unsigned char)a_g;
     S.slot_mdc[base + cand] = (unsigned char)mdc;
@@ -936,7 +1033,7 @@ __device__ int rl_elastic_preempt_all(Ctx& c, int d, double now) {
     double f = S.s_fused[base + k];
     double T = d_unit_time(n, f, c.lc3(d, 1));
     double done = S.s_done[base + k] +
-                  fmax(0.0, now - S.s_start[base + k]) / fmax(T, 1e-300);
+                  fmax(0.0, now - S.s_lastupd[base + k]) / fmax(T, 1e-300);
     double size = S.s_size[base + k];
     done = fmin(size, done);
     int64_t pb = (int64_t)c.r * S.pp_cap + count;
@@ -980,22 +1077,22 @@ __device__ int rl_elastic_preempt_all(Ctx& c, int d, double now) {
 // (reference :849-890); returns true if a request was issued (pause)
 __device__ bool rl_try_drain_request(Ctx& c, int d, double now) {
   const EngineDesc& S = *c.S;
-  double qsize;
+  double qsize, qenq;
   float qnetlat;
   int qjid, qing;
   bool popped = false;
   int from_inf = 0;
   if (c.free_gpus(d) > 0) {
-    if (S.inf_priority && queue_pop(c, d, 0, qsize, qnetlat, qjid, qing)) {
+    if (S.inf_priority && queue_pop(c, d, 0, qsize, qnetlat, qjid, qing, qenq)) {
       popped = true;
       from_inf = 1;
-    } else if (queue_pop(c, d, 1, qsize, qnetlat, qjid, qing)) {
+    } else if (queue_pop(c, d, 1, qsize, qnetlat, qjid, qing, qenq)) {
       popped = true;
     }
   }
   if (popped)
     rl_request(c, PEND_DRAIN, now, from_inf ? 0 : 1, qing, qsize, qnetlat,
-               qjid, d, from_inf);
+               qjid, d, from_inf, qenq);
   return popped;
 }
 
@@ -1005,7 +1102,7 @@ __device__ void rl_request_realloc(Ctx& c, double now) {
   int cur = S.pp_cursor[c.r];
   int64_t pb = (int64_t)c.r * S.pp_cap + cur;
   rl_request(c, PEND_REALLOC, now, 1, S.pp_ing[pb], S.pp_size[pb],
-             S.pp_netlat[pb], S.pp_jid[pb], S.pp_dc[pb], 0);
+             S.pp_netlat[pb], S.pp_jid[pb], S.pp_dc[pb], 0, now);
 }
 
 // ---------------- the advance kernel ----------------
@@ -1147,13 +1244,16 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int d_tgt = a_dc;
       if (c.free_gpus(d_tgt) <= 0) {
         // no free target: push the job back at the FRONT of its source queue
-        // (reference :858-861)
-        queue_push_front(c, src_d, from_inf ? 0 : 1, size, netlat, jid, ing);
+        // (reference :858-861), keeping its original enqueue time
+        queue_push_front(c, src_d, from_inf ? 0 : 1, size, netlat, jid, ing,
+                         S.pend_enq[c.r]);
       } else {
         int n_sel = max(1, min(min(a_g + 1, c.free_gpus(d_tgt)), S.max_gpj));
         double f = rl_energy_freq(c, d_tgt, jt, n_sel);
         rl_start_job(c, d_tgt, jt, size, netlat, jid, ing, n_sel, f, c.now,
                      s0, a_dc, a_g, mdc, mg, n_sel);
+        if (lane == 0)
+          S.sum_wait[c.r] += fmax(0.0, c.now - S.pend_enq[c.r]);
       }
     } else {  // PEND_REALLOC: resume the pool entry at the cursor on its DC
       int cur = S.pp_cursor[c.r];
@@ -1163,7 +1263,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         // no free GPUs: re-queue on the training queue (oracle's fix of
         // reference Appendix A.6 job-stranding)
         queue_push(c, d_src, 1, S.pp_size[pb], S.pp_netlat[pb],
-                   S.pp_jid[pb], S.pp_ing[pb]);
+                   S.pp_jid[pb], S.pp_ing[pb], c.now);
       } else {
         int n_rl = max(1, min(min(a_g + 1, c.free_gpus(d_src)), S.max_gpj));
         double f = rl_energy_freq(c, d_src, 1, n_rl);
@@ -1314,7 +1414,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         if (lane == 0 && !S.trace_mode) c.hs->arr_next[idx] = t_min + ia_rl;
         lds_fence();
         rl_request(c, PEND_ARRIVAL, t_min, jt, ing, size, 0.0f, jid,
-                   -1, 0);
+                   -1, 0, t_min);
         paused = true;
         break;
       }
@@ -1462,7 +1562,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       } else {
         // queueing drops the RL trace: a later chsac drain assigns a fresh
         // observation/action (reference :849-889 overwrites them)
-        queue_push(c, d, jt, size, netlat, jid, ing);
+        queue_push(c, d, jt, size, netlat, jid, ing, t_min);
       }
 
     } else if (kind == 2) {
@@ -1492,7 +1592,6 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         // metrics
         S.jobs_done[c.r] += 1;
         S.sum_lat[c.r] += t_min - start;
-        S.sum_wait[c.r] += 0.0;
         if (jt == 0) {
           S.jobs_done_inf[c.r] += 1;
           S.sum_lat_inf[c.r] += t_min - start;
@@ -1698,6 +1797,7 @@ class BatchedSimHip {
     T_PTR(p_active, double); T_PTR(sum_tpt, double); T_PTR(n_running, int);
     T_PTR(dc_min_finish, double); T_PTR(dc_min_slot, int);
     T_PTR(s_finish, double); T_PTR(s_start, double); T_PTR(s_size, double);
+    T_PTR(s_lastupd, double); T_PTR(s_seq, int); T_PTR(seq_ctr, int);
     T_PTR(s_fused, double); T_PTR(s_netlat, float); T_PTR(s_jid, int);
     S_.s_gpus = reinterpret_cast<short*>(t_["s_gpus"].data_ptr<int16_t>());
     S_.s_jtype = reinterpret_cast<char*>(t_["s_jtype"].data_ptr<int8_t>());
@@ -1710,6 +1810,8 @@ class BatchedSimHip {
     S_.x_jtype = reinterpret_cast<char*>(t_["x_jtype"].data_ptr<int8_t>());
     S_.x_ing = reinterpret_cast<char*>(t_["x_ing"].data_ptr<int8_t>());
     T_PTR(q_head, int); T_PTR(q_len, int); T_PTR(q_size, double);
+    T_PTR(q_enq, double);
+    if (S_.algo == A_CAP_GREEDY) T_PTR(snap_f, double);
     T_PTR(q_netlat, float); T_PTR(q_jid, int);
     S_.q_ing = reinterpret_cast<char*>(t_["q_ing"].data_ptr<int8_t>());
     T_PTR(b_n, int); T_PTR(b_s, double);
@@ -1743,7 +1845,7 @@ class BatchedSimHip {
       T_PTR(req_mg, int); T_PTR(resp_dc, int); T_PTR(resp_g, int);
       T_PTR(pend_kind, int); T_PTR(pend_size, double); T_PTR(pend_netlat, float);
       T_PTR(pend_jid, int); T_PTR(pend_ing, int); T_PTR(pend_jt, int);
-      T_PTR(pend_dc, int); T_PTR(pend_from_inf, int);
+      T_PTR(pend_dc, int); T_PTR(pend_from_inf, int); T_PTR(pend_enq, double);
       T_PTR(slot_s0, float);
       S_.slot_adc = reinterpret_cast<unsigned char*>(t_["slot_adc"].data_ptr<uint8_t>());
       S_.slot_ag = reinterpret_cast<unsigned char*>(t_["slot_ag"].data_ptr<uint8_t>());
