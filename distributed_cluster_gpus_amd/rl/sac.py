@@ -17,6 +17,7 @@ from typing import Callable, Dict, Optional
 
 import torch
 import torch.nn as nn
+import torch.nn.functional as F
 
 from .masking import sample_categorical
 from .networks import HybridActor, StateEncoder, TwinQuantileCritic
@@ -24,13 +25,19 @@ from .networks import HybridActor, StateEncoder, TwinQuantileCritic
 
 def quantile_huber_loss(pred: torch.Tensor, target: torch.Tensor,
                         taus: torch.Tensor) -> torch.Tensor:
-    """QR-DQN quantile Huber loss; pred/target [B, N], taus [N]."""
-    delta = target.unsqueeze(2) - pred.unsqueeze(1)          # [B, N_tgt, N_pred]
-    abs_delta = torch.abs(delta)
-    huber = torch.where(abs_delta <= 1.0, 0.5 * delta ** 2, abs_delta - 0.5)
-    tau = taus.view(1, -1, 1)
-    weight = torch.abs((delta.detach() < 0).float() - tau)
-    return (weight * huber).mean()
+    """QR quantile-Huber loss (pred/target [B, N], taus [N]).
+
+    Built from the standard decomposition: a kappa=1 Huber on every
+    (target_j, pred_i) pair — smooth_l1 with beta=1 IS that Huber — weighted
+    by the asymmetric pinball factor |tau - 1{u < 0}| on the undershoot
+    indicator of u = target - pred (spec: the reference's QR critic update,
+    simcore/rl/hybrid_sac.py:83-93)."""
+    pred_pairs = pred.unsqueeze(1).expand(-1, target.shape[1], -1)
+    tgt_pairs = target.unsqueeze(2).expand_as(pred_pairs)    # [B, N_tgt, N_pred]
+    huber = F.smooth_l1_loss(pred_pairs, tgt_pairs, reduction="none", beta=1.0)
+    undershoot = (tgt_pairs.detach() < pred_pairs.detach()).float()
+    pinball = (taus.view(1, -1, 1) - undershoot).abs()
+    return (pinball * huber).mean()
 
 
 class MaskedHybridSAC(nn.Module):
