@@ -62,6 +62,8 @@ class BatchedEngine:
                  rl_warmup: int = 1000, rl_buffer: int = 200000,
                  rl_train_interval: int = 256, rl_agent=None,
                  rl_stats_interval: int = 100,
+                 rl_serve: str = "device", rl_deterministic: bool = False,
+                 rl_tr_limit: Optional[int] = None,
                  tr_cap: int = 262144, arrival_trace=None,
                  subwave: int = 64, **_unused):
         if algo not in ALGOS:
@@ -250,8 +252,14 @@ class BatchedEngine:
         self._rl_warmup = int(rl_warmup)
         self._rl_train_interval = int(rl_train_interval)
         self._rl_stats_interval = int(rl_stats_interval)
-        self._rl_det = False  # deterministic (greedy) host serving: parity mode
+        self._rl_det = bool(rl_deterministic)
         self.rl_updates = 0
+        # serving mode: "device" = the actor MLP + masked Gumbel-max sampling
+        # run INSIDE the advance kernel from a flat weights buffer (replicas
+        # never pause for a host policy round-trip — the round-1 206k ev/s
+        # bottleneck); "host" = pause/resume with a batched torch forward
+        # (used for parity testing and injected non-standard agents)
+        self._serve_device = self.is_rl and rl_serve == "device"
         if self.is_rl:
             t["req_flag"] = torch.zeros(R, **i32)
             t["req_obs"] = torch.zeros((R, obs_dim), **f32)
@@ -339,6 +347,22 @@ class BatchedEngine:
                                      n_dc=n_dc,
                                      n_g=int(scenario.policy.max_gpus_per_job),
                                      device=str(dev), seed=seed)
+            # flat fp32 actor-weights buffer for in-kernel serving
+            # (layout documented in replica_engine.hip EngineDesc.pw)
+            n_g = int(scenario.policy.max_gpus_per_job)
+            H = 256
+            if self._serve_device and not self._actor_is_servable(H):
+                import warnings
+                warnings.warn("injected agent has non-standard actor dims; "
+                              "falling back to host policy serving")
+                self._serve_device = False
+            wsize = (obs_dim * H + H) + 2 * (H * H + H) + (H * H + H) + \
+                (H * n_dc + n_dc) + (H * H + H) + (H * n_g + n_g)
+            t["policy_weights"] = torch.zeros(wsize if self._serve_device
+                                              else 1, **f32)
+            self._rl_hid = H
+            self._tr_limit = int(rl_tr_limit) if rl_tr_limit is not None else \
+                min(int(tr_cap) // 2, self._rl_train_interval * 32)
         self.t = t
         self.arrival_inf, self.arrival_trn = arrival_inf, arrival_trn
 
@@ -371,9 +395,16 @@ class BatchedEngine:
             "pp_cap": getattr(self, "_pp_cap", 0),
             "trace_mode": int(self.trace_mode),
             "trace_cap": getattr(self, "trace_cap", 0),
+            "serve_device": int(getattr(self, "_serve_device", False)),
+            "rl_hid": getattr(self, "_rl_hid", 256),
+            "rl_det": int(self._rl_det) if self.is_rl else 0,
+            "tr_limit": getattr(self, "_tr_limit", 0),
         }
         self._sim = self._mod.BatchedSimHip(t, cfg)
         self.meter = ThroughputMeter()
+        if self._serve_device:
+            self._build_weight_refs()
+            self._refresh_policy_weights()
 
     @staticmethod
     def _mode_id(mode: str) -> int:
@@ -604,6 +635,54 @@ class BatchedEngine:
                 # sync-free eager SAC step (no stats, tensorized PID)
                 self.rl.train_step(self.replay.sample(self._rl_batch),
                                    compute_stats=False)
+        if self._serve_device:
+            self._refresh_policy_weights()
+
+    # ---- in-kernel serving support ----
+    def _actor_lins(self):
+        """The 7 Linear layers of the served actor, in buffer-layout order."""
+        rl = self.rl
+        enc = rl.encoder.net
+        return [enc[0], enc[2], enc[4],
+                rl.actor.head_dc[0], rl.actor.head_dc[2],
+                rl.actor.head_g[0], rl.actor.head_g[2]]
+
+    def _actor_is_servable(self, H: int) -> bool:
+        try:
+            lins = self._actor_lins()
+        except (AttributeError, IndexError, TypeError):
+            return False
+        shapes = [tuple(l.weight.shape) for l in lins]
+        n_dc = self.sc.n_dc
+        n_g = int(self.sc.policy.max_gpus_per_job)
+        want = [(H, self.obs_dim), (H, H), (H, H),
+                (H, H), (n_dc, H), (H, H), (n_g, H)]
+        return shapes == want and n_dc <= 8 and n_g <= 8 and self.obs_dim <= 64
+
+    def _build_weight_refs(self):
+        """Precompute (buffer-slice view, source-parameter) pairs so each
+        refresh is a handful of small device-to-device copies."""
+        buf = self.t["policy_weights"]
+        pairs = []
+        off = 0
+        for lin in self._actor_lins():
+            w = lin.weight          # [out, in] -> stored transposed [in][out]
+            n = w.numel()
+            pairs.append((buf[off:off + n].view(w.shape[1], w.shape[0]), w))
+            off += n
+            b = lin.bias
+            pairs.append((buf[off:off + b.numel()], b))
+            off += b.numel()
+        assert off == buf.numel(), (off, buf.numel())
+        self._w_pairs = pairs
+
+    @torch.no_grad()
+    def _refresh_policy_weights(self):
+        """Push the current actor weights into the kernel's serving buffer
+        (called after every train round; bounds policy staleness together
+        with the kernel's tr_limit launch yield)."""
+        for dst, src in self._w_pairs:
+            dst.copy_(src.t() if dst.dim() == 2 else src)
 
     def _rl_service(self, n_req=None, n_tr=None):
         """Back-compat single-rank service entry (serve + ingest + local-

@@ -178,6 +178,21 @@ struct EngineDesc {
   // ===== CHSAC-AF (RL-in-the-loop) state; null unless algo == A_CHSAC =====
   int obs_dim;                    // 1 + 6*n_dc
   double sla_p99_ms;
+  // --- device-side policy serving (serve_device=1): the actor MLP runs
+  // INSIDE the advance kernel from a flat weights buffer the host refreshes
+  // after each SAC train round, so replicas never pause for a host policy
+  // round-trip (round-1 bottleneck: 206k ev/s host-loop-bound).  Layout of
+  // pw (all fp32, [in][out]-major so the out index is lane-coalesced):
+  //   enc1 Wt[obs_dim][hid] b[hid] | enc2 Wt[hid][hid] b | enc3 Wt[hid][hid] b
+  //   | hdc1 Wt[hid][hid] b | hdc2 Wt[hid][n_dc] b[n_dc]
+  //   | hg1 Wt[hid][hid] b | hg2 Wt[hid][n_g] b[n_g]
+  int serve_device;               // 0 = host pause/resume, 1 = in-kernel
+  int hid;                        // encoder/head hidden width (<= RL_MAX_HID)
+  int n_g;                        // g-head size (max_gpj)
+  int rl_det;                     // 1 = greedy argmax (no Gumbel draw)
+  int tr_limit;                   // yield the launch when the transition ring
+                                  //   reaches this (bounds policy staleness)
+  const float* pw;                // flat policy weights
   // action request/response, one slot per replica
   int* req_flag;                  // [r] ReqFlag
   float* req_obs;                 // [r][obs_dim]
@@ -298,6 +313,12 @@ struct Ctx {
   Hot* hs;      // LDS-resident hot state of this wave's replica
   double* l_fin;  // LDS mirror of this replica's s_finish[total_slots]
   double* l_xt;   // LDS mirror of this replica's x_time[tcap]
+  // chsac serve-device scratch (LDS): obs vector, two activation ping-pong
+  // buffers, and the head logits (dc at +0, g at +8)
+  float* l_obs;
+  float* l_act_a;
+  float* l_act_b;
+  float* l_logits;
   int r;        // local replica index
   int lane;
   PhiloxState rng;
@@ -915,6 +936,101 @@ __device__ double rl_energy_freq(Ctx& c, int d, int jt, int n) {
                           S.n_freq, n);
 }
 
+// ---------------- device-side actor forward + sampling ----------------
+// One wave evaluates the full CHSAC actor for ONE observation:
+// enc (obs->hid->hid->hid, ReLU) then the two categorical heads
+// (hid->hid->n_dc / n_g).  Weights are [in][out]-major, so at every input
+// index the SUBW lanes read consecutive floats (fully coalesced, L2-resident
+// at ~1.1 MB total); activations live in LDS.  ~280k FMA per call.
+constexpr int RL_MAX_HID = 256;
+constexpr int RL_MAX_OBS = 64;
+constexpr int RL_MAX_LOG = 16;
+
+__device__ void rl_dense(const float* x, int in, const float* Wt,
+                         const float* b, int out, float* y, int lane,
+                         bool relu) {
+  for (int j = lane; j < out; j += SUBW) {
+    float acc = 0.0f;
+    for (int i = 0; i < in; ++i) acc = fmaf(x[i], Wt[i * out + j], acc);
+    acc += b[j];
+    y[j] = relu ? fmaxf(acc, 0.0f) : acc;
+  }
+  lds_fence();
+}
+
+// logits for both heads; obs in LDS, dc logits at L[0..n_dc), g at L[8..)
+__device__ void rl_logits_impl(const float* pw, int D, int H, int n_dc,
+                               int n_g, const float* obs, float* A, float* B,
+                               float* L, int lane) {
+  const float *w1 = pw, *b1 = w1 + D * H;
+  const float *w2 = b1 + H, *b2 = w2 + H * H;
+  const float *w3 = b2 + H, *b3 = w3 + H * H;
+  const float *hdc1 = b3 + H, *hdc1b = hdc1 + H * H;
+  const float *hdc2 = hdc1b + H, *hdc2b = hdc2 + H * n_dc;
+  const float *hg1 = hdc2b + n_dc, *hg1b = hg1 + H * H;
+  const float *hg2 = hg1b + H, *hg2b = hg2 + H * n_g;
+  rl_dense(obs, D, w1, b1, H, A, lane, true);
+  rl_dense(A, H, w2, b2, H, B, lane, true);
+  rl_dense(B, H, w3, b3, H, A, lane, true);        // h3 stays in A
+  rl_dense(A, H, hdc1, hdc1b, H, B, lane, true);
+  rl_dense(B, H, hdc2, hdc2b, n_dc, L, lane, false);
+  rl_dense(A, H, hg1, hg1b, H, B, lane, true);
+  rl_dense(B, H, hg2, hg2b, n_g, L + 8, lane, false);
+}
+
+__device__ void rl_actor_logits(Ctx& c) {
+  const EngineDesc& S = *c.S;
+  rl_logits_impl(S.pw, S.obs_dim, S.hid, S.n_dc, S.n_g, c.l_obs,
+                 c.l_act_a, c.l_act_b, c.l_logits, c.lane);
+}
+
+// direct (key, ctr) uniform — distinct per-category Gumbel draws without
+// advancing the per-replica stream n times serially
+__device__ __forceinline__ double u01_at(uint64_t key, uint64_t ctr) {
+  uint32_t w[4];
+  philox4x32(key, ctr, w);
+  return ((w[0] >> 5) * 67108864.0 + (w[1] >> 6)) *
+         (1.0 / 9007199254740992.0);
+}
+
+// masked Gumbel-max categorical pick (equivalent in distribution to the host
+// path's log_softmax+Gumbel argmax: the log-softmax shift cancels in the
+// argmax).  mask==0 falls back to all-valid, mirroring the host guard.
+// Wave-uniform: every lane computes the same winner from the same draws.
+__device__ int rl_pick(Ctx& c, const float* logits, int mask, int n) {
+  const EngineDesc& S = *c.S;
+  if (mask == 0) mask = (1 << n) - 1;
+  double best = -D_INF;
+  int arg = 0;
+  if (S.rl_det) {
+    for (int j = 0; j < n; ++j) {
+      double z = logits[j];
+      if ((mask >> j & 1) && z > best) { best = z; arg = j; }
+    }
+    return arg;
+  }
+  uint64_t ctr0 = c.rng.ctr;
+  c.rng.ctr += n;
+  for (int j = 0; j < n; ++j) {
+    if (!(mask >> j & 1)) continue;
+    double u = fmax(u01_at(c.rng.key, ctr0 + j), 1e-20);
+    double z = (double)logits[j] - log(-log(u));
+    if (z > best) { best = z; arg = j; }
+  }
+  return arg;
+}
+
+// build obs+masks into LDS, run the actor, sample (a_dc, a_g)
+__device__ void rl_serve_inline(Ctx& c, double now, int& a_dc, int& a_g,
+                                int& mdc, int& mg) {
+  const EngineDesc& S = *c.S;
+  rl_build_obs(c, now, c.l_obs);
+  rl_build_masks(c, mdc, mg);
+  rl_actor_logits(c);
+  a_dc = rl_pick(c, c.l_logits, mdc, S.n_dc);
+  a_g = rl_pick(c, c.l_logits + 8, mg, S.n_g);
+}
+
 // start a job carrying an RL trace; returns chosen slot via start_job's path.
 // (duplicates start_job, then fills the rl trace of the slot just used)
 __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
@@ -1105,6 +1221,128 @@ __device__ void rl_request_realloc(Ctx& c, double now) {
              S.pp_netlat[pb], S.pp_jid[pb], S.pp_dc[pb], 0, now);
 }
 
+// ---- action EXECUTION bodies, shared by the host resume path and the
+// in-kernel serving path ----
+
+// complete an RL-routed arrival: push the WAN transfer carrying the trace
+// (reference :570-588 with the stashed action)
+__device__ void rl_do_arrival(Ctx& c, double now, int jt, int ing,
+                              double size, int jid, int a_dc, int a_g,
+                              const float* s0, int mdc, int mg) {
+  const EngineDesc& S = *c.S;
+  int d_sel = a_dc;
+  int n_sel = a_g + 1;
+  double lnet = S.wan_lat[ing * S.n_dc + d_sel];
+  double bw = S.wan_bw[ing * S.n_dc + d_sel];
+  double xfer = bw > 0.0 ? S.payload_gb[jt] / bw : 0.0;
+  int cand = INT_MAX;
+  for (int k = c.lane; k < S.tcap; k += SUBW) {
+    if (c.l_xt[k] >= D_INF) { cand = k; break; }
+  }
+#pragma unroll
+  for (int off = SUBW / 2; off > 0; off >>= 1)
+    cand = min(cand, __shfl_xor(cand, off, 64));
+  if (cand == INT_MAX) {
+    if (c.lane == 0) atomicOr(&S.err[c.r], ERR_XFER_OVF);
+    return;
+  }
+  int64_t at = (int64_t)c.r * S.tcap + cand;
+  for (int k = c.lane; k < S.obs_dim; k += SUBW)
+    S.x_s0[at * S.obs_dim + k] = s0[k];
+  if (c.lane == 0) {
+    c.l_xt[cand] = now + lnet + xfer;
+    S.x_size[at] = size;
+    S.x_netlat[at] = (float)lnet;
+    S.x_jid[at] = jid;
+    S.x_dc[at] = (char)d_sel;
+    S.x_jtype[at] = (char)jt;
+    S.x_ing[at] = (char)ing;
+    S.x_nsel[at] = (short)n_sel;
+    S.x_adc[at] = (unsigned char)a_dc;
+    S.x_ag[at] = (unsigned char)a_g;
+    S.x_mdc[at] = (unsigned char)mdc;
+    S.x_mg[at] = (unsigned char)mg;
+    S.x_has_rl[at] = 1;
+  }
+  store_fence();
+}
+
+// execute a drain action on the popped job: start on the RL's target DC, or
+// push back at the FRONT of the source queue when the target has no free
+// GPUs (reference :849-890)
+__device__ void rl_do_drain_action(Ctx& c, double now, int src_d,
+                                   int from_inf, int jt, int ing, double size,
+                                   float netlat, int jid, double enq,
+                                   int a_dc, int a_g, const float* s0,
+                                   int mdc, int mg) {
+  const EngineDesc& S = *c.S;
+  int d_tgt = a_dc;
+  if (c.free_gpus(d_tgt) <= 0) {
+    queue_push_front(c, src_d, from_inf ? 0 : 1, size, netlat, jid, ing, enq);
+  } else {
+    int n_sel = max(1, min(min(a_g + 1, c.free_gpus(d_tgt)), S.max_gpj));
+    double f = rl_energy_freq(c, d_tgt, jt, n_sel);
+    rl_start_job(c, d_tgt, jt, size, netlat, jid, ing, n_sel, f, now,
+                 s0, a_dc, a_g, mdc, mg, n_sel);
+    if (c.lane == 0) S.sum_wait[c.r] += fmax(0.0, now - enq);
+  }
+}
+
+// in-kernel one-job queue drain (serve_device): pop, serve, execute — the
+// replica continues its event loop with no host round-trip
+__device__ void rl_drain_inline(Ctx& c, int d, double now) {
+  const EngineDesc& S = *c.S;
+  double qsize, qenq;
+  float qnetlat;
+  int qjid, qing;
+  int from_inf = 0;
+  bool popped = false;
+  if (c.free_gpus(d) > 0) {
+    if (S.inf_priority && queue_pop(c, d, 0, qsize, qnetlat, qjid, qing, qenq)) {
+      popped = true;
+      from_inf = 1;
+    } else if (queue_pop(c, d, 1, qsize, qnetlat, qjid, qing, qenq)) {
+      popped = true;
+    }
+  }
+  if (!popped) return;
+  int a_dc, a_g, mdc, mg;
+  rl_serve_inline(c, now, a_dc, a_g, mdc, mg);
+  rl_do_drain_action(c, now, d, from_inf, from_inf ? 0 : 1, qing, qsize,
+                     qnetlat, qjid, qenq, a_dc, a_g, c.l_obs, mdc, mg);
+}
+
+// in-kernel elastic reallocation chain (serve_device): fresh obs + actor
+// forward per pool entry, all within this launch (reference
+// _rl_reallocate_training_jobs :498-534; failed resumes re-queue)
+__device__ void rl_realloc_inline(Ctx& c, double now) {
+  const EngineDesc& S = *c.S;
+  int cnt = S.pp_count[c.r];
+  for (int cur = 0; cur < cnt; ++cur) {
+    int a_dc, a_g, mdc, mg;
+    rl_serve_inline(c, now, a_dc, a_g, mdc, mg);
+    int64_t pb = (int64_t)c.r * S.pp_cap + cur;
+    int d_src = S.pp_dc[pb];
+    if (c.free_gpus(d_src) <= 0) {
+      queue_push(c, d_src, 1, S.pp_size[pb], S.pp_netlat[pb],
+                 S.pp_jid[pb], S.pp_ing[pb], now);
+    } else {
+      int n_rl = max(1, min(min(a_g + 1, c.free_gpus(d_src)), S.max_gpj));
+      double f = rl_energy_freq(c, d_src, 1, n_rl);
+      rl_start_job(c, d_src, 1, S.pp_size[pb], S.pp_netlat[pb],
+                   S.pp_jid[pb], S.pp_ing[pb], n_rl, f, now,
+                   &S.pp_s0[pb * S.obs_dim], S.pp_adc[pb], S.pp_ag[pb],
+                   mdc, mg, n_rl, S.pp_done[pb], S.pp_pcount[pb],
+                   S.pp_has_rl[pb]);
+    }
+  }
+  if (c.lane == 0) {
+    S.pp_count[c.r] = 0;
+    S.pp_cursor[c.r] = 0;
+  }
+  store_fence();
+}
+
 // ---------------- the advance kernel ----------------
 template <int ALGO>
 __global__ void __launch_bounds__(THREADS_PER_BLOCK)
@@ -1144,7 +1382,9 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     size_t hot_sz = (sizeof(Hot) + 15) & ~size_t(15);
     size_t fin_sz = SUBW == 64 ? (size_t)S.total_slots * sizeof(double) : 0;
     size_t xt_sz = SUBW == 64 ? (size_t)S.tcap * sizeof(double) : 0;
-    size_t stride = hot_sz + fin_sz + xt_sz;
+    size_t rl_sz = ALGO == A_CHSAC
+        ? (RL_MAX_OBS + 2 * RL_MAX_HID + RL_MAX_LOG) * sizeof(float) : 0;
+    size_t stride = hot_sz + fin_sz + xt_sz + rl_sz;
     char* base = smem + (threadIdx.x / SUBW) * stride;
     c.hs = reinterpret_cast<Hot*>(base);
     if (SUBW == 64) {
@@ -1153,6 +1393,13 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     } else {
       c.l_fin = S.s_finish + sbase;
       c.l_xt = S.x_time + (int64_t)c.r * S.tcap;
+    }
+    if (ALGO == A_CHSAC) {
+      float* rb = reinterpret_cast<float*>(base + hot_sz + fin_sz + xt_sz);
+      c.l_obs = rb;
+      c.l_act_a = rb + RL_MAX_OBS;
+      c.l_act_b = c.l_act_a + RL_MAX_HID;
+      c.l_logits = c.l_act_b + RL_MAX_HID;
     }
   }
   {
@@ -1203,58 +1450,12 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     int jid = S.pend_jid[c.r];
     if (pk == PEND_ARRIVAL) {
       // complete the arrival: RL chose (dc, g); push the WAN transfer
-      int d_sel = a_dc;
-      int n_sel = a_g + 1;
-      double lnet = S.wan_lat[ing * S.n_dc + d_sel];
-      double bw = S.wan_bw[ing * S.n_dc + d_sel];
-      double xfer = bw > 0.0 ? S.payload_gb[jt] / bw : 0.0;
-      int cand = INT_MAX;
-      for (int k = lane; k < S.tcap; k += SUBW) {
-        if (c.l_xt[k] >= D_INF) { cand = k; break; }
-      }
-#pragma unroll
-      for (int off = SUBW / 2; off > 0; off >>= 1)
-        cand = min(cand, __shfl_xor(cand, off, 64));
-      if (cand == INT_MAX) {
-        if (lane == 0) atomicOr(&S.err[c.r], ERR_XFER_OVF);
-      } else {
-        int64_t at = (int64_t)c.r * S.tcap + cand;
-        for (int k = lane; k < S.obs_dim; k += SUBW)
-          S.x_s0[at * S.obs_dim + k] = s0[k];
-        if (lane == 0) {
-          c.l_xt[cand] = c.now + lnet + xfer;
-          S.x_size[at] = size;
-          S.x_netlat[at] = (float)lnet;
-          S.x_jid[at] = jid;
-          S.x_dc[at] = (char)d_sel;
-          S.x_jtype[at] = (char)jt;
-          S.x_ing[at] = (char)ing;
-          S.x_nsel[at] = (short)n_sel;
-          S.x_adc[at] = (unsigned char)a_dc;
-          S.x_ag[at] = (unsigned char)a_g;
-          S.x_mdc[at] = (unsigned char)mdc;
-          S.x_mg[at] = (unsigned char)mg;
-          S.x_has_rl[at] = 1;
-        }
-      }
+      rl_do_arrival(c, c.now, jt, ing, size, jid, a_dc, a_g, s0, mdc, mg);
     } else if (pk == PEND_DRAIN) {
       // one queued job, RL chose a target DC + g
-      int src_d = S.pend_dc[c.r];
-      int from_inf = S.pend_from_inf[c.r];
-      int d_tgt = a_dc;
-      if (c.free_gpus(d_tgt) <= 0) {
-        // no free target: push the job back at the FRONT of its source queue
-        // (reference :858-861), keeping its original enqueue time
-        queue_push_front(c, src_d, from_inf ? 0 : 1, size, netlat, jid, ing,
-                         S.pend_enq[c.r]);
-      } else {
-        int n_sel = max(1, min(min(a_g + 1, c.free_gpus(d_tgt)), S.max_gpj));
-        double f = rl_energy_freq(c, d_tgt, jt, n_sel);
-        rl_start_job(c, d_tgt, jt, size, netlat, jid, ing, n_sel, f, c.now,
-                     s0, a_dc, a_g, mdc, mg, n_sel);
-        if (lane == 0)
-          S.sum_wait[c.r] += fmax(0.0, c.now - S.pend_enq[c.r]);
-      }
+      rl_do_drain_action(c, c.now, S.pend_dc[c.r], S.pend_from_inf[c.r], jt,
+                         ing, size, netlat, jid, S.pend_enq[c.r],
+                         a_dc, a_g, s0, mdc, mg);
     } else {  // PEND_REALLOC: resume the pool entry at the cursor on its DC
       int cur = S.pp_cursor[c.r];
       int64_t pb = (int64_t)c.r * S.pp_cap + cur;
@@ -1304,6 +1505,12 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   }
 
   while (!skip_loop && n_events < max_ev) {
+    // serve-device staleness bound: yield the launch once the transition
+    // ring is full enough for a host train round (checked every 32 events;
+    // the read races benignly with other replicas' atomicAdds)
+    if (ALGO == A_CHSAC && S.serve_device && S.tr_limit > 0 &&
+        (n_events & 31) == 0 && *S.tr_count >= S.tr_limit)
+      break;
     // ---- 1. next event: wave argmin over candidate sources ----
     // per-lane candidate: value + kind/idx
     double v = D_INF;
@@ -1413,11 +1620,19 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         }
         if (lane == 0 && !S.trace_mode) c.hs->arr_next[idx] = t_min + ia_rl;
         lds_fence();
-        rl_request(c, PEND_ARRIVAL, t_min, jt, ing, size, 0.0f, jid,
-                   -1, 0, t_min);
-        paused = true;
-        break;
-      }
+        if (S.serve_device) {
+          // in-kernel policy: forward + sample + execute, no pause
+          int a_dc, a_g, mdc, mg;
+          rl_serve_inline(c, t_min, a_dc, a_g, mdc, mg);
+          rl_do_arrival(c, t_min, jt, ing, size, jid, a_dc, a_g, c.l_obs,
+                        mdc, mg);
+        } else {
+          rl_request(c, PEND_ARRIVAL, t_min, jt, ing, size, 0.0f, jid,
+                     -1, 0, t_min);
+          paused = true;
+          break;
+        }
+      } else {
       // routing
       int d_sel;
       if (S.trace_mode && trace_d >= 0) {
@@ -1533,6 +1748,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       }
       if (lane == 0 && !S.trace_mode) c.hs->arr_next[idx] = t_min + ia;
       lds_fence();
+      }  // end non-chsac arrival path
 
     } else if (kind == 1) {
       // ===== WAN transfer complete: admission =====
@@ -1651,13 +1867,19 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           if (n_train > 1) {
             int cnt = rl_elastic_preempt_all(c, d, t_min);
             if (cnt > 0) {
-              rl_request_realloc(c, t_min);
-              paused = true;
-              break;
+              if (S.serve_device) {
+                rl_realloc_inline(c, t_min);  // whole chain, this launch
+              } else {
+                rl_request_realloc(c, t_min);
+                paused = true;
+                break;
+              }
             }
           }
         }
-        if (rl_try_drain_request(c, d, t_min)) {
+        if (S.serve_device) {
+          rl_drain_inline(c, d, t_min);
+        } else if (rl_try_drain_request(c, d, t_min)) {
           paused = true;
           break;
         }
@@ -1732,6 +1954,56 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     S.rng_ctr[c.r] = c.rng.ctr;
     S.ev_count[c.r] += n_events;
   }
+}
+
+// standalone batched actor forward (test/verification path): one subwave
+// slot per observation row, same device math as the in-engine serving
+__global__ void __launch_bounds__(THREADS_PER_BLOCK)
+rl_forward_kernel(const float* pw, const float* obs, int B, int obs_dim,
+                  int hid, int n_dc, int n_g, float* out_dc, float* out_g) {
+  int slot = (blockIdx.x * blockDim.x + threadIdx.x) / SUBW;
+  int lane = threadIdx.x & (SUBW - 1);
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  size_t per = (RL_MAX_OBS + 2 * RL_MAX_HID + RL_MAX_LOG) * sizeof(float);
+  float* rb = reinterpret_cast<float*>(smem + (threadIdx.x / SUBW) * per);
+  if (slot >= B) return;
+  float* l_obs = rb;
+  float* A = rb + RL_MAX_OBS;
+  float* Bv = A + RL_MAX_HID;
+  float* L = Bv + RL_MAX_HID;
+  for (int k = lane; k < obs_dim; k += SUBW)
+    l_obs[k] = obs[(int64_t)slot * obs_dim + k];
+  lds_fence();
+  rl_logits_impl(pw, obs_dim, hid, n_dc, n_g, l_obs, A, Bv, L, lane);
+  for (int k = lane; k < n_dc; k += SUBW)
+    out_dc[(int64_t)slot * n_dc + k] = L[k];
+  for (int k = lane; k < n_g; k += SUBW)
+    out_g[(int64_t)slot * n_g + k] = L[8 + k];
+}
+
+std::vector<torch::Tensor> rl_forward_debug(torch::Tensor pw,
+                                            torch::Tensor obs,
+                                            int64_t hid, int64_t n_dc,
+                                            int64_t n_g) {
+  TORCH_CHECK(pw.is_cuda() && obs.is_cuda() && pw.dtype() == torch::kFloat32);
+  int B = obs.size(0), D = obs.size(1);
+  auto out_dc = torch::empty({B, n_dc}, obs.options());
+  auto out_g = torch::empty({B, n_g}, obs.options());
+  int rpb = REPLICAS_PER_BLOCK;
+  int blocks = (B + rpb - 1) / rpb;
+  size_t shmem = (size_t)rpb *
+      (RL_MAX_OBS + 2 * RL_MAX_HID + RL_MAX_LOG) * sizeof(float);
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(rl_forward_kernel, dim3(blocks), dim3(THREADS_PER_BLOCK),
+                     shmem, stream, pw.data_ptr<float>(),
+                     obs.data_ptr<float>(), B, D, (int)hid, (int)n_dc,
+                     (int)n_g, out_dc.data_ptr<float>(),
+                     out_g.data_ptr<float>());
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess)
+    throw std::runtime_error(std::string("rl_forward_kernel: ") +
+                             hipGetErrorString(e));
+  return {out_dc, out_g};
 }
 
 // ---------------- host side ----------------
@@ -1840,6 +2112,18 @@ class BatchedSimHip {
     S_.tr_cap = cfg.contains("tr_cap") ? cfg["tr_cap"].cast<int>() : 0;
     S_.elastic = cfg.contains("elastic") ? cfg["elastic"].cast<int>() : 0;
     S_.pp_cap = cfg.contains("pp_cap") ? cfg["pp_cap"].cast<int>() : 0;
+    S_.serve_device = cfg.contains("serve_device") ? cfg["serve_device"].cast<int>() : 0;
+    S_.hid = cfg.contains("rl_hid") ? cfg["rl_hid"].cast<int>() : 256;
+    S_.n_g = S_.max_gpj;
+    S_.rl_det = cfg.contains("rl_det") ? cfg["rl_det"].cast<int>() : 0;
+    S_.tr_limit = cfg.contains("tr_limit") ? cfg["tr_limit"].cast<int>() : 0;
+    if (S_.serve_device) {
+      if (S_.hid > RL_MAX_HID || S_.obs_dim > RL_MAX_OBS ||
+          S_.n_dc > 8 || S_.n_g > 8)
+        throw std::runtime_error("serve_device limits: hid<=256, obs<=64, "
+                                 "n_dc<=8, n_g<=8");
+      S_.pw = t_["policy_weights"].data_ptr<float>();
+    }
     if (S_.algo == A_CHSAC) {
       T_PTR(req_flag, int); T_PTR(req_obs, float); T_PTR(req_mdc, int);
       T_PTR(req_mg, int); T_PTR(resp_dc, int); T_PTR(resp_g, int);
@@ -1889,12 +2173,15 @@ class BatchedSimHip {
     int rpb = REPLICAS_PER_BLOCK;
     int blocks = (S_.n_rep + rpb - 1) / rpb;
     dim3 grid(blocks), block(THREADS_PER_BLOCK);
-    // dynamic LDS: per replica, Hot (+ s_finish/x_time mirrors at SUBW==64)
+    // dynamic LDS: per replica, Hot (+ s_finish/x_time mirrors at SUBW==64,
+    // + actor-forward scratch for chsac)
     size_t hot_sz = (sizeof(Hot) + 15) & ~size_t(15);
     size_t per_rep = hot_sz;
     if (SUBW == 64)
       per_rep += (size_t)S_.total_slots * sizeof(double) +
                  (size_t)S_.tcap * sizeof(double);
+    if (S_.algo == A_CHSAC)
+      per_rep += (RL_MAX_OBS + 2 * RL_MAX_HID + RL_MAX_LOG) * sizeof(float);
     size_t shmem = rpb * per_rep;
     if (shmem > 64 * 1024)
       throw std::runtime_error(
@@ -1947,4 +2234,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def(py::init<py::dict, py::dict>())
       .def("advance", &dcg::BatchedSimHip::advance,
            py::arg("t_target"), py::arg("max_events"));
+  m.def("rl_forward_debug", &dcg::rl_forward_debug,
+        "batched actor forward via the in-kernel serving math "
+        "(pw, obs[B,D], hid, n_dc, n_g) -> (logits_dc, logits_g)",
+        py::arg("pw"), py::arg("obs"), py::arg("hid"), py::arg("n_dc"),
+        py::arg("n_g"));
 }

@@ -297,7 +297,7 @@ def test_chsac_batched_respects_masks():
     eng = BatchedEngine(sc, inf, trn, algo="chsac_af", replicas=16,
                         duration=60.0, log_interval=5.0, out_dir=None,
                         seed=3, enable_logs=False, rl_warmup=10**9,
-                        events_per_launch=5000)
+                        events_per_launch=5000, rl_serve="host")
     checked = {"n": 0}
     inner = eng.rl.select_action_batch
 
@@ -604,3 +604,95 @@ def test_rccl_collectives_smoke():
         torch.cuda.synchronize()
     finally:
         dist.destroy_process_group()
+
+
+@needs_gpu
+def test_device_actor_forward_matches_torch():
+    """The in-kernel actor forward (ops rl_forward_debug — the same device
+    math the serve_device path runs) must match the torch actor's logits on
+    random obs within fp32 accumulation tolerance."""
+    from distributed_cluster_gpus_amd.ops import load_sim_hip
+    from distributed_cluster_gpus_amd.rl.agent import (CHSACAgent,
+                                                       CHSACAgentConfig)
+    mod = load_sim_hip()
+    torch.manual_seed(11)
+    obs_dim, n_dc, n_g, H = 49, 8, 8, 256
+    agent = CHSACAgent(CHSACAgentConfig(
+        obs_dim=obs_dim, n_dc=n_dc, n_g_choices=n_g,
+        constraints={"latency_p99": 500.0}, device="cuda"))
+    # flat buffer with the engine's layout
+    segs = []
+    enc = agent.encoder.net
+    for lin in (enc[0], enc[2], enc[4], agent.actor.head_dc[0],
+                agent.actor.head_dc[2], agent.actor.head_g[0],
+                agent.actor.head_g[2]):
+        segs.append(lin.weight.detach().t().contiguous().reshape(-1))
+        segs.append(lin.bias.detach().reshape(-1))
+    pw = torch.cat(segs).float().cuda()
+    obs = torch.randn(256, obs_dim, device="cuda") * 100.0
+    out_dc, out_g = mod.rl_forward_debug(pw, obs, H, n_dc, n_g)
+    with torch.no_grad():
+        ref_dc, ref_g = agent.actor(agent.encoder(obs))
+    for got, ref in ((out_dc, ref_dc), (out_g, ref_g)):
+        diff = (got - ref).abs().max().item()
+        scale = ref.abs().max().item() + 1.0
+        assert diff <= 2e-4 * scale, f"logit mismatch {diff} (scale {scale})"
+
+
+@needs_gpu
+def test_chsac_device_serving_runs_and_trains():
+    """serve_device end-to-end: the actor runs inside the advance kernel,
+    transitions stream out, SAC trains, weights refresh — and throughput per
+    launch is far beyond the pause/resume path's one-event-per-pause."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="poisson", rate=2.0)
+    trn = ArrivalProcess(mode="poisson", rate=0.3)
+    eng = BatchedEngine(sc, inf, trn, algo="chsac_af", replicas=64,
+                        duration=200.0, log_interval=5.0, out_dir=None,
+                        seed=7, enable_logs=False,
+                        rl_warmup=256, rl_batch=64, rl_train_interval=64,
+                        events_per_launch=50000)
+    assert eng._serve_device
+    st = eng.run()
+    assert st["jobs_completed"] > 0
+    assert eng.replay.size > 0
+    assert eng.rl_updates > 0, "SAC never trained in device-serve mode"
+    assert int(eng.t["err"].max().item()) == 0
+    b = eng.replay.sample(32)
+    assert torch.isfinite(b["r"]).all()
+    assert (b["costs"]["latency_p99"] >= 0).all()
+    eng.validate_state()
+    # device mode must finish in FAR fewer launches than events processed
+    # (pause/resume needed ~one launch per decision)
+    assert eng.timing["launches"] * 100 < st["events"]
+
+
+@needs_gpu
+def test_chsac_device_vs_host_serving_consistent():
+    """Device and host serving are the same process statistically: with the
+    same seed and workload, job counts and energies land within a loose
+    population band (different RNG streams for action sampling)."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+
+    def run(mode):
+        torch.manual_seed(5)
+        sc = paper_scenario()
+        inf = ArrivalProcess(mode="poisson", rate=2.0)
+        trn = ArrivalProcess(mode="poisson", rate=0.2)
+        eng = BatchedEngine(sc, inf, trn, algo="chsac_af", replicas=48,
+                            duration=120.0, log_interval=5.0, out_dir=None,
+                            seed=9, enable_logs=False, rl_warmup=10**9,
+                            rl_serve=mode, events_per_launch=20000)
+        st = eng.run()
+        return (st["jobs_completed"] / 48.0,
+                float(eng.t["energy_j"].sum().item()) / 48.0)
+
+    jobs_d, en_d = run("device")
+    jobs_h, en_h = run("host")
+    assert abs(jobs_d - jobs_h) / max(jobs_h, 1) < 0.05
+    assert abs(en_d - en_h) / en_h < 0.05
