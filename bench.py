@@ -93,6 +93,53 @@ def measure_rl_steps_per_sec(device, n_updates=200, batch=256):
     return n_updates / (time.perf_counter() - t0)
 
 
+def measure_rl_loop(device, wall_budget_s=8.0, replicas=4096):
+    """CHSAC-AF RL-IN-THE-LOOP throughput (BASELINE config 4's shape): the
+    batched engine with in-kernel policy serving, transitions streaming into
+    replay and SAC training at the production cadence.  Reference baseline:
+    127 events/s (BASELINE.md chsac_af run).  Untimed region of the bench."""
+    import time as _t
+
+    import torch
+
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="sinusoid", rate=6.0, amp=0.6, period=300.0)
+    trn = ArrivalProcess(mode="poisson", rate=0.3)
+    eng = BatchedEngine(sc, inf, trn, algo="chsac_af", replicas=replicas,
+                        duration=1e9, log_interval=20.0, out_dir=None,
+                        seed=1, enable_logs=False, rl_warmup=2048,
+                        rl_batch=256, rl_train_interval=256,
+                        rl_stats_interval=0, events_per_launch=100000)
+    t = eng.t
+
+    def cycle():
+        eng._sim.advance(eng.end_time, eng.events_per_launch)
+        status = torch.stack([t["err"].max(), t["tr_count"][0]]).cpu()
+        assert int(status[0]) == 0, "engine error in rl bench"
+        n_new = eng._rl_ingest(n_tr=int(status[1]))
+        eng._tr_backlog = getattr(eng, "_tr_backlog", 0) + n_new
+        steps = 0
+        if eng.replay.size >= eng._rl_warmup:
+            steps = min(64, eng._tr_backlog // eng._rl_train_interval)
+            eng._tr_backlog -= steps * eng._rl_train_interval
+        eng._rl_train(steps)
+
+    cycle()  # warmup (captures the hipGraph train step lazily)
+    torch.cuda.synchronize(device)
+    ev0 = int(t["ev_count"].sum().item())
+    up0 = eng.rl_updates
+    t0 = _t.perf_counter()
+    while _t.perf_counter() - t0 < wall_budget_s:
+        cycle()
+    torch.cuda.synchronize(device)
+    el = _t.perf_counter() - t0
+    return ((int(t["ev_count"].sum().item()) - ev0) / el,
+            (eng.rl_updates - up0) / el)
+
+
 def main():
     args = parse_args()
     import torch
@@ -169,8 +216,10 @@ def main():
     value = events / elapsed
 
     rl_steps_s = None
+    rl_loop_ev_s = rl_loop_up_s = None
     if args.with_rl and rank == 0:
         rl_steps_s = measure_rl_steps_per_sec(device)
+        rl_loop_ev_s, rl_loop_up_s = measure_rl_loop(device)
 
     if rank == 0:
         baseline = 4003.0  # BASELINE.md reference events/sec (CPU, 1 core)
@@ -199,6 +248,9 @@ def main():
                 "topology": "8 DC / 1488 GPUs / 8 ingresses",
                 "rl_train_steps_per_sec": rl_steps_s,
                 "rl_baseline_steps_per_sec": 36.9,
+                "rl_loop_events_per_sec": rl_loop_ev_s,
+                "rl_loop_updates_per_sec": rl_loop_up_s,
+                "rl_loop_baseline_events_per_sec": 127.0,
             },
         }
         print(json.dumps(out))
