@@ -63,6 +63,7 @@ class BatchedEngine:
                  rl_train_interval: int = 256, rl_agent=None,
                  rl_stats_interval: int = 100,
                  rl_serve: str = "device", rl_deterministic: bool = False,
+                 rl_exact_p99: bool = False,
                  rl_tr_limit: Optional[int] = None,
                  tr_cap: int = 262144, arrival_trace=None,
                  subwave: int = 64, **_unused):
@@ -313,6 +314,13 @@ class BatchedEngine:
             t["lat_hist"] = torch.zeros((R, 2, 64), **i32)
             t["lat_count"] = torch.zeros((R, 2), **i64)
             t["lat_sum"] = torch.zeros((R, 2), **f64)
+            # exact sliding-window p99 (parity mode): sorted window + ring
+            # reproducing the reference's np.percentile over 2048 sojourns
+            self._exact_p99 = bool(rl_exact_p99)
+            self._p99_win = 2048
+            p99_shape = (R, 2, self._p99_win) if self._exact_p99 else (1, 1, 1)
+            t["p99_sorted"] = torch.zeros(p99_shape, **f64)
+            t["p99_ring"] = torch.zeros(p99_shape, **f64)
             t["tr_count"] = torch.zeros(1, **i32)
             t["tr_s0"] = torch.zeros((tr_cap, obs_dim), **f32)
             t["tr_s1"] = torch.zeros((tr_cap, obs_dim), **f32)
@@ -399,6 +407,8 @@ class BatchedEngine:
             "rl_hid": getattr(self, "_rl_hid", 256),
             "rl_det": int(self._rl_det) if self.is_rl else 0,
             "tr_limit": getattr(self, "_tr_limit", 0),
+            "exact_p99": int(getattr(self, "_exact_p99", False)),
+            "p99_win": getattr(self, "_p99_win", 2048),
         }
         self._sim = self._mod.BatchedSimHip(t, cfg)
         self.meter = ThroughputMeter()
@@ -577,11 +587,28 @@ class BatchedEngine:
         # guard: a fully-false mask wedges the categorical; allow all
         m_dc[m_dc.sum(dim=1) == 0] = True
         m_g[m_g.sum(dim=1) == 0] = True
-        with torch.no_grad():
-            a = self.rl.select_action_batch(obs, m_dc, m_g,
-                                            deterministic=self._rl_det)
-        t["resp_dc"][idx] = a["dc"].to(torch.int32)
-        t["resp_g"][idx] = a["g"].to(torch.int32)
+        if self._rl_det:
+            # parity mode: serve one request at a time through the SAME
+            # scalar entry the oracle uses, so batch-size-dependent GEMM
+            # reduction order can never flip a near-tie argmax
+            obs_np = obs.cpu().numpy()
+            mdc_np = m_dc.cpu().numpy()
+            mg_np = m_g.cpu().numpy()
+            dcs, gs = [], []
+            for i in range(obs_np.shape[0]):
+                a = self.rl.select_action(obs_np[i], mdc_np[i], mg_np[i],
+                                          deterministic=True)
+                dcs.append(a["dc"])
+                gs.append(a["g"])
+            t["resp_dc"][idx] = torch.as_tensor(dcs, dtype=torch.int32,
+                                                device=self.device)
+            t["resp_g"][idx] = torch.as_tensor(gs, dtype=torch.int32,
+                                               device=self.device)
+        else:
+            with torch.no_grad():
+                a = self.rl.select_action_batch(obs, m_dc, m_g)
+            t["resp_dc"][idx] = a["dc"].to(torch.int32)
+            t["resp_g"][idx] = a["g"].to(torch.int32)
         t["req_flag"][idx] = 2  # REQ_READY
 
     def _rl_ingest(self, n_tr=None) -> int:

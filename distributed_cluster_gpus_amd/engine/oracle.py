@@ -64,7 +64,7 @@ class OracleEngine:
                  rl_buffer: int = 200000, rl_agent=None,
                  logger=None, show_progress: bool = False,
                  cluster_writer=None, job_writer=None,
-                 arrival_recorder=None):
+                 arrival_recorder=None, rl_deterministic: bool = False):
         if algo not in ALGOS:
             raise ValueError(f"unknown algo {algo!r}")
         self.sc = scenario
@@ -88,6 +88,9 @@ class OracleEngine:
         # optional list collecting (t, ingress, jtype, size) per arrival —
         # used to build exact-replay traces for the GPU engine
         self.arrival_recorder = arrival_recorder
+        # parity mode: greedy (argmax) policy actions so a frozen agent's
+        # trajectory is reproducible across engines
+        self.rl_deterministic = bool(rl_deterministic)
         self.elastic_scaling = bool(elastic_scaling) and algo == "chsac_af"
 
         # CPython Mersenne stream: seeded exactly like the reference
@@ -325,7 +328,7 @@ class OracleEngine:
         elif self.algo == "chsac_af" and self.rl is not None:
             obs = self._rl_obs()
             m_dc, m_g = self._rl_masks()
-            a = self.rl.select_action(obs, m_dc, m_g, deterministic=False)
+            a = self.rl.select_action(obs, m_dc, m_g, deterministic=self.rl_deterministic)
             dc_name = self._dc_names[int(a["dc"])]
             n_sel = int(a["g"]) + 1
             job.rl_state0 = obs
@@ -675,7 +678,7 @@ class OracleEngine:
             job = pre.job
             obs = self._rl_obs()
             m_dc, m_g = self._rl_masks()
-            a = self.rl.select_action(obs, m_dc, m_g, deterministic=False)
+            a = self.rl.select_action(obs, m_dc, m_g, deterministic=self.rl_deterministic)
             n_rl = max(1, min(int(a["g"]) + 1, dc.free_gpus,
                               self.sc.policy.max_gpus_per_job))
             f_rl = self._energy_freq_with_deadline(dc, job, n_rl)
@@ -700,7 +703,7 @@ class OracleEngine:
             if self.algo == "chsac_af" and self.rl is not None:
                 obs = self._rl_obs()
                 m_dc, m_g = self._rl_masks()
-                a = self.rl.select_action(obs, m_dc, m_g, deterministic=False)
+                a = self.rl.select_action(obs, m_dc, m_g, deterministic=self.rl_deterministic)
                 dc_tgt = self.dcs[self._dc_names[int(a["dc"])]]
                 if dc_tgt.free_gpus <= 0:
                     (dc.q_inf if nxt.jtype == "inference" else dc.q_train).insert(0, nxt)

@@ -248,6 +248,15 @@ struct EngineDesc {
   int* lat_hist;                  // [r][2][LAT_BINS]
   long long* lat_count;           // [r][2]
   double* lat_sum;                // [r][2]
+  // exact sliding-window p99 (parity mode): the reference computes an exact
+  // percentile over the last 2048 sojourns (simulator_paper_multi.py:728-737)
+  // where the fast path approximates from the log histogram.  exact_p99=1
+  // maintains a SORTED window + an insertion-order ring per (r, jtype) and
+  // reproduces np.percentile(buf, 99) bit-for-bit.
+  int exact_p99;
+  int p99_win;                    // window size (2048)
+  double* p99_sorted;             // [r][2][win]
+  double* p99_ring;               // [r][2][win]
   // transition ring (global across replicas; host drains between launches)
   int tr_cap;
   int* tr_count;                  // [1] atomicAdd cursor
@@ -844,6 +853,19 @@ __device__ void rl_build_obs(Ctx& c, double now, float* out) {
 __device__ double rl_p99_ms(Ctx& c, int jt) {
   const EngineDesc& S = *c.S;
   long long total = S.lat_count[c.r * 2 + jt];
+  if (S.exact_p99) {
+    // np.percentile(buf, 99) over the sorted window: linear interpolation
+    // at virtual index 0.99*(n-1)
+    long long n = total < S.p99_win ? total : S.p99_win;
+    if (n < 5) return -1.0;
+    const double* a = &S.p99_sorted[(int64_t)(c.r * 2 + jt) * S.p99_win];
+    double vi = 0.99 * (double)(n - 1);
+    long long lo = (long long)vi;
+    double frac = vi - (double)lo;
+    double v = a[lo];
+    if (lo + 1 < n) v = v + (a[lo + 1] - v) * frac;
+    return v * 1000.0;
+  }
   if (total < 5) return -1.0;
   long long target = (long long)(0.99 * (double)total);
   long long cum = 0;
@@ -866,8 +888,29 @@ __device__ void rl_record_latency(Ctx& c, int jt, double sojourn_s) {
   int b = (int)((l + 4.0) / 8.0 * LAT_BINS);
   b = max(0, min(LAT_BINS - 1, b));
   S.lat_hist[(c.r * 2 + jt) * LAT_BINS + b] += 1;
-  S.lat_count[c.r * 2 + jt] += 1;
+  long long cnt = S.lat_count[c.r * 2 + jt];
+  S.lat_count[c.r * 2 + jt] = cnt + 1;
   S.lat_sum[c.r * 2 + jt] += sojourn_s;
+  if (S.exact_p99) {
+    // maintain the sorted window (serial on lane 0 — parity-mode only)
+    const int W = S.p99_win;
+    double* srt = &S.p99_sorted[(int64_t)(c.r * 2 + jt) * W];
+    double* ring = &S.p99_ring[(int64_t)(c.r * 2 + jt) * W];
+    long long n = cnt < W ? cnt : W;
+    if (cnt >= W) {
+      // evict the oldest: find its slot in the sorted array, close the gap
+      double old = ring[cnt % W];
+      int p = 0;
+      while (p < n && srt[p] < old) ++p;   // first slot holding `old`
+      for (int q = p; q + 1 < n; ++q) srt[q] = srt[q + 1];
+      n -= 1;
+    }
+    int p = 0;
+    while (p < n && srt[p] < sojourn_s) ++p;
+    for (int q = (int)n; q > p; --q) srt[q] = srt[q - 1];
+    srt[p] = sojourn_s;
+    ring[cnt % W] = sojourn_s;
+  }
 }
 
 // masks (reference _upgr_masks :1055-1082): DC valid if it has free GPUs;
@@ -2158,6 +2201,12 @@ class BatchedSimHip {
       T_PTR(lat_hist, int);
       S_.lat_count = reinterpret_cast<long long*>(t_["lat_count"].data_ptr<int64_t>());
       T_PTR(lat_sum, double);
+      S_.exact_p99 = cfg.contains("exact_p99") ? cfg["exact_p99"].cast<int>() : 0;
+      S_.p99_win = cfg.contains("p99_win") ? cfg["p99_win"].cast<int>() : 2048;
+      if (S_.exact_p99) {
+        T_PTR(p99_sorted, double);
+        T_PTR(p99_ring, double);
+      }
       T_PTR(tr_count, int);
       T_PTR(tr_s0, float); T_PTR(tr_s1, float);
       S_.tr_adc = reinterpret_cast<unsigned char*>(t_["tr_adc"].data_ptr<uint8_t>());

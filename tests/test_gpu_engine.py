@@ -696,3 +696,95 @@ def test_chsac_device_vs_host_serving_consistent():
     jobs_h, en_h = run("host")
     assert abs(jobs_d - jobs_h) / max(jobs_h, 1) < 0.05
     assert abs(en_d - en_h) / en_h < 0.05
+
+
+@needs_gpu
+def test_chsac_pinned_policy_parity(tmp_path):
+    """with a FROZEN agent served greedily (deterministic),
+    the exact-p99 window enabled, and the oracle's recorded arrival trace,
+    one GPU replica reproduces the oracle's chsac trajectory event-for-event
+    — job log identical, and the recorded transitions (rewards, p99/gpu_over
+    costs, masks) match, which pins the engine's RL mechanics including the
+    exact percentile path (the fast path's histogram p99 stays the
+    documented approximation)."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.engine.oracle import OracleEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    from distributed_cluster_gpus_amd.rl.agent import (CHSACAgent,
+                                                       CHSACAgentConfig)
+    torch.manual_seed(42)
+    duration = 150.0
+    agent = CHSACAgent(CHSACAgentConfig(
+        obs_dim=49, n_dc=8, n_g_choices=8,
+        constraints={"latency_p99": 500.0, "gpu_over": 0.0}, device="cuda"))
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="poisson", rate=2.5)
+    trn = ArrivalProcess(mode="poisson", rate=0.3)
+    rec = []
+    out_o = str(tmp_path / "oracle")
+    o_eng = OracleEngine(sc, inf, trn, algo="chsac_af", duration=duration,
+                         log_interval=5.0, out_dir=out_o, seed=123,
+                         rl_agent=agent, rl_warmup=10**9, rl_device="cuda",
+                         rl_deterministic=True, arrival_recorder=rec)
+    o_eng.run()
+
+    ing_idx = {n: i for i, n in enumerate(sc.ingress_names)}
+    NS = sc.n_ing * 2
+    streams = [[] for _ in range(NS)]
+    for (tt, ing, jtype, size, dc) in rec:
+        s_id = ing_idx[ing] * 2 + (0 if jtype == "inference" else 1)
+        streams[s_id].append((tt, size))
+    cap = max(len(x) for x in streams) + 1
+    times = np.full((1, NS, cap), 1e300)
+    sizes = np.zeros((1, NS, cap), np.float64)
+    for s_id, entries in enumerate(streams):
+        for k, (tt, size) in enumerate(entries):
+            times[0, s_id, k] = tt
+            sizes[0, s_id, k] = size
+
+    sc2 = paper_scenario()
+    out_g = str(tmp_path / "gpu")
+    eng = BatchedEngine(sc2, inf, trn, algo="chsac_af", replicas=1,
+                        duration=duration, log_interval=5.0, out_dir=out_g,
+                        seed=999, enable_logs=True, rl_agent=agent,
+                        rl_warmup=10**9, rl_serve="host",
+                        rl_deterministic=True, rl_exact_p99=True,
+                        arrival_trace=(times, sizes))
+    eng.run()
+
+    import pandas as pd
+    jo = pd.read_csv(os.path.join(out_o, "job_log.csv"))
+    jg = pd.read_csv(os.path.join(out_g, "job_log.csv"))
+    assert len(jo) == len(jg), f"job count {len(jo)} vs {len(jg)}"
+    jo = jo.sort_values("jid").reset_index(drop=True)
+    jg = jg.sort_values("jid").reset_index(drop=True)
+    for col in ("jid", "ingress", "type", "dc", "n_gpus"):
+        assert (jo[col] == jg[col]).all(), f"column {col} diverged"
+    for col, tol in (("size", 1e-9), ("f_used", 1e-12),
+                     ("start_s", 1e-9), ("finish_s", 1e-9),
+                     ("latency_s", 1e-9), ("preempt_count", 0)):
+        d = (jo[col] - jg[col]).abs().max()
+        assert d <= tol, f"column {col} max diff {d}"
+
+    # transition-stream parity: same count, same actions/rewards/costs/masks
+    # (costs carry the EXACT sliding-window p99 -> this pins the exact-p99
+    # device path against the reference's np.percentile)
+    orep, grep_ = o_eng.replay, eng.replay
+    assert orep.size == grep_.size > 100, (orep.size, grep_.size)
+    n = orep.size
+    assert torch.equal(orep.a_dc[:n].cpu(), grep_.a_dc[:n].cpu())
+    assert torch.equal(orep.a_g[:n].cpu(), grep_.a_g[:n].cpu())
+    assert torch.allclose(orep.r[:n].cpu(), grep_.r[:n].cpu(),
+                          atol=1e-6, rtol=1e-5)
+    oc = {name: orep.costs[:n, k].cpu()
+          for k, name in enumerate(orep.cost_names)}
+    gc = {name: grep_.costs[:n, k].cpu()
+          for k, name in enumerate(grep_.cost_names)}
+    assert torch.allclose(oc["latency_p99"], gc["latency_p99"],
+                          atol=1e-4, rtol=1e-5), "exact-p99 cost diverged"
+    assert torch.equal(oc["gpu_over"], gc["gpu_over"])
+    assert torch.equal(orep.mask_dc[:n].cpu(), grep_.mask_dc[:n].cpu())
+    assert torch.equal(orep.mask_g[:n].cpu(), grep_.mask_g[:n].cpu()), \
+        "g-mask diverged (p99-vs-SLA cap path)"
+    assert torch.allclose(orep.s[:n].cpu(), grep_.s[:n].cpu(), atol=1e-4)
