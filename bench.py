@@ -40,6 +40,8 @@ def parse_args():
     p.add_argument("--algo", type=str, default="default_policy")
     p.add_argument("--seed", type=int, default=123)
     p.add_argument("--subwave", type=int, default=64, choices=[8, 64])
+    p.add_argument("--fp32-coeff", type=int, default=0,
+                   help="opt-in fp32 decision-score eval in the sim kernels")
     p.add_argument("--with-rl", type=int, default=1,
                    help="also measure CHSAC-AF SAC updates/sec (untimed region)")
     return p.parse_args()
@@ -176,7 +178,8 @@ def main():
                         replicas=total_replicas, duration=duration,
                         log_interval=5.0, out_dir=None, seed=args.seed,
                         device=device, rank=rank, world=world, qcap=qcap,
-                        enable_logs=False, subwave=args.subwave)
+                        enable_logs=False, subwave=args.subwave,
+                        fp32_coeff_eval=bool(args.fp32_coeff))
 
     ev = eng.t["ev_count"]
 
@@ -245,6 +248,7 @@ def main():
                 "replicas_per_gpu": total_replicas // world,
                 "events_per_step": args.events_per_step,
                 "arrivals": "sinusoid inf 6/s amp 0.6 period 300 + poisson trn 0.3/s",
+                "fp32_coeff_eval": bool(args.fp32_coeff),
                 "topology": "8 DC / 1488 GPUs / 8 ingresses",
                 "rl_train_steps_per_sec": rl_steps_s,
                 "rl_baseline_steps_per_sec": 36.9,

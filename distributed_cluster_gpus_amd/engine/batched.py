@@ -63,7 +63,7 @@ class BatchedEngine:
                  rl_train_interval: int = 256, rl_agent=None,
                  rl_stats_interval: int = 100,
                  rl_serve: str = "device", rl_deterministic: bool = False,
-                 rl_exact_p99: bool = False,
+                 rl_exact_p99: bool = False, fp32_coeff_eval: bool = False,
                  rl_tr_limit: Optional[int] = None,
                  tr_cap: int = 262144, arrival_trace=None,
                  subwave: int = 64, **_unused):
@@ -387,6 +387,7 @@ class BatchedEngine:
             "dvfs_high": float(scenario.policy.dvfs_high),
             "power_cap": float(power_cap),
             "eco_obj": _ECO_IDS[eco_objective],
+            "fp32_score": int(bool(fp32_coeff_eval)),
             "num_fixed": int(num_fixed_gpus),
             "fixed_freq": float(fixed_freq) if fixed_freq else 0.0,
             "payload_inf_gb": PAYLOAD_GB[0], "payload_trn_gb": PAYLOAD_GB[1],

@@ -79,6 +79,8 @@ struct EngineDesc {
   int algo, max_gpj, inf_priority, scale_out_low, energy_aware;
   double dvfs_low, dvfs_high, power_cap;
   int eco_obj, num_fixed;
+  int fp32_score;                 // opt-in fp32 decision-score eval (grid /
+                                  //   energy-freq argmins; times stay f64)
   double fixed_freq;
   double payload_gb[2];
   // arrival processes by jtype: mode 0=poisson 1=sinusoid 2=off
@@ -450,8 +452,11 @@ __device__ void decide_nf(Ctx& c, int d, int jt, float size, double now,
   const EngineDesc& S = *c.S;
   int free = c.free_gpus(d);
   if (ALGO == A_JOINT_NF) {
-    GridPick g = wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
-                                  S.n_freq, S.max_gpj, 0, 0.0, 0.0, false, 0.0);
+    GridPick g = S.fp32_score
+        ? wave_grid_argmin_f32(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                               S.n_freq, S.max_gpj, 0, 0.0, 0.0, false, 0.0)
+        : wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                           S.n_freq, S.max_gpj, 0, 0.0, 0.0, false, 0.0);
     n_out = g.n; f_out = g.f;
   } else if (ALGO == A_BANDIT) {
     // UCB1 (learners.py:20-36): uniform-redundant serial loop over arms
@@ -474,17 +479,22 @@ __device__ void decide_nf(Ctx& c, int d, int jt, float size, double now,
     f_out = bf;
   } else if (ALGO == A_CARBON_COST) {
     double price = c.price_kwh(now);
-    GridPick g = (price > 0.0)
-        ? wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
-                           S.n_freq, S.max_gpj, 2, 0.0, price, false, 0.0)
+    int obj = price > 0.0 ? 2 : 1;
+    double ci = price > 0.0 ? 0.0 : S.carbon[d];
+    GridPick g = S.fp32_score
+        ? wave_grid_argmin_f32(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                               S.n_freq, S.max_gpj, obj, ci, price, false, 0.0)
         : wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
-                           S.n_freq, S.max_gpj, 1, S.carbon[d], 0.0, false, 0.0);
+                           S.n_freq, S.max_gpj, obj, ci, price, false, 0.0);
     n_out = g.n; f_out = g.f;
   } else if (ALGO == A_DEBUG) {
     n_out = S.num_fixed;
     f_out = S.fixed_freq > 0 ? S.fixed_freq
-            : wave_energy_freq(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
-                               S.n_freq, S.num_fixed);
+            : (S.fp32_score
+               ? wave_energy_freq_f32(c.pc3(d, jt), c.lc3(d, jt),
+                                      S.freq_levels, S.n_freq, S.num_fixed)
+               : wave_energy_freq(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                                  S.n_freq, S.num_fixed));
   } else {  // heuristic family
     n_out = heuristic_alloc(c, d, jt);
     f_out = c.hs->cur_freq[d];
@@ -589,9 +599,13 @@ __device__ void drain_queues(Ctx& c, int d, double now) {
       // the reference's drain path always uses the CARBON objective for
       // carbon_cost (simulator_paper_multi.py:909-920), unlike its admission
       // path which prefers cost when a price is set (:622-645)
-      GridPick g = wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
-                                    S.n_freq, S.max_gpj, 1, S.carbon[d], 0.0,
-                                    false, 0.0);
+      GridPick g = S.fp32_score
+          ? wave_grid_argmin_f32(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                                 S.n_freq, S.max_gpj, 1, S.carbon[d], 0.0,
+                                 false, 0.0)
+          : wave_grid_argmin(c.pc3(d, jt), c.lc3(d, jt), S.freq_levels,
+                             S.n_freq, S.max_gpj, 1, S.carbon[d], 0.0,
+                             false, 0.0);
       n = g.n; f = g.f;
     } else if (ALGO == A_JOINT_NF || ALGO == A_BANDIT) {
       // explicit drain branches in the reference (:892-907); everything else
@@ -2079,6 +2093,7 @@ class BatchedSimHip {
     S_.dvfs_high = cfg["dvfs_high"].cast<double>();
     S_.power_cap = cfg["power_cap"].cast<double>();
     S_.eco_obj = cfg["eco_obj"].cast<int>();
+    S_.fp32_score = cfg.contains("fp32_score") ? cfg["fp32_score"].cast<int>() : 0;
     S_.num_fixed = cfg["num_fixed"].cast<int>();
     S_.fixed_freq = cfg["fixed_freq"].cast<double>();
     S_.payload_gb[0] = cfg["payload_inf_gb"].cast<double>();

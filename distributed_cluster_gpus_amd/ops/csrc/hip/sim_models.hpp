@@ -145,4 +145,76 @@ __device__ __forceinline__ double wave_energy_freq(
   return freq_levels[wk];
 }
 
+// ---- fp32 DECISION-SCORE variants (opt-in, BASELINE config 5's "fp16/fp32
+// coeff eval") ----
+// Only the candidate-ranking math drops to fp32; the chosen (n, f) feeds the
+// usual f64 time/energy models, so event times and energy integrals keep
+// full precision.  A decision can differ from the f64 path only when two
+// candidates score within fp32 rounding of each other (documented,
+// measurable divergence — off by default).
+__device__ __forceinline__ float f_gpu_power(float f, const double* c3) {
+  f = fmaxf(0.0f, f);
+  return (float)c3[0] * f * f * f + (float)c3[1] * f + (float)c3[2];
+}
+__device__ __forceinline__ float f_unit_time(int n, float f, const double* c3) {
+  n = max(1, n);
+  f = fmaxf(1e-9f, f);
+  if (n == 1) return (float)c3[0] + (float)c3[1] / f;
+  return ((float)c3[0] + (float)c3[1] / f + (float)c3[2] * n) / n;
+}
+
+__device__ __forceinline__ GridPick wave_grid_argmin_f32(
+    const double* pc3, const double* lc3, const double* freq_levels,
+    int n_freq, int n_max, int objective, double ci, double price,
+    bool has_ddl, double ddl) {
+  int lane = sub_lane();
+  int n_cand = n_max * n_freq;
+  float best_sc = 3.0e38f;
+  int best_ci = INT_MAX;
+  for (int cand = lane; cand < n_cand; cand += SUBW) {
+    int n = cand / n_freq + 1;
+    float f = (float)freq_levels[cand % n_freq];
+    float T = f_unit_time(n, f, lc3);
+    float E = (float)max(0, n) * f_gpu_power(f, pc3) * T;
+    if (has_ddl && T > (float)ddl) continue;
+    float sc = E;
+    if (objective == 1) sc = E * (float)ci;
+    else if (objective == 2) sc = (E / 3.6e6f) * (float)price;
+    if (sc < best_sc) { best_sc = sc; best_ci = cand; }
+  }
+  double v;
+  int wi;
+  wave_argmin_idx_f64((double)best_sc, best_ci, v, wi);
+  GridPick out;
+  out.found = v < 3.0e38;
+  if (!out.found) {
+    out.n = 1; out.f = 0; out.T = 0; out.P = 0; out.E = 0;
+    return out;
+  }
+  // full-precision models on the CHOSEN candidate
+  out.n = wi / n_freq + 1;
+  out.f = freq_levels[wi % n_freq];
+  out.T = d_unit_time(out.n, out.f, lc3);
+  out.P = d_job_power(out.n, out.f, pc3);
+  out.E = out.P * out.T;
+  return out;
+}
+
+__device__ __forceinline__ double wave_energy_freq_f32(
+    const double* pc3, const double* lc3, const double* freq_levels,
+    int n_freq, int n) {
+  int lane = sub_lane();
+  float best_sc = 3.0e38f;
+  int best_k = INT_MAX;
+  for (int k = lane; k < n_freq; k += SUBW) {
+    float f = (float)freq_levels[k];
+    float sc = (float)max(0, n) * f_gpu_power(f, pc3) * f_unit_time(n, f, lc3);
+    if (sc < best_sc) { best_sc = sc; best_k = k; }
+  }
+  double v;
+  int wk;
+  wave_argmin_idx_f64((double)best_sc, best_k, v, wk);
+  return freq_levels[wk];
+}
+
 }  // namespace dcg
