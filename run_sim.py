@@ -84,6 +84,17 @@ def parse_args(argv=None):
                         "batched = MI355X HIP replica engine.")
     p.add_argument("--replicas", type=int, default=4096,
                    help="Monte-Carlo replicas (batched engine only).")
+    p.add_argument("--rl-serve", type=str, default="device",
+                   choices=["device", "host"],
+                   help="chsac_af policy serving on the batched engine: "
+                        "in-kernel actor (device) or host pause/resume.")
+    p.add_argument("--rl-exact-p99", action="store_true", default=False,
+                   help="Exact 2048-sample sliding-window p99 on the batched "
+                        "engine (parity mode) instead of the histogram "
+                        "approximation.")
+    p.add_argument("--fp32-coeff-eval", action="store_true", default=False,
+                   help="fp32 decision-score evaluation in the sim kernels "
+                        "(batched engine; times/energies stay f64).")
     p.add_argument("--single-dc", action="store_true", default=False,
                    help="Use the single-DC debug topology.")
     p.add_argument("--rl-checkpoint", type=str, default=None,
@@ -152,7 +163,9 @@ def main(argv=None):
     elif args.engine == "batched":
         from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
         eng = BatchedEngine(sc, arrival_inf, arrival_trn,
-                            replicas=args.replicas, **common)
+                            replicas=args.replicas, rl_serve=args.rl_serve,
+                            rl_exact_p99=args.rl_exact_p99,
+                            fp32_coeff_eval=args.fp32_coeff_eval, **common)
     else:
         from distributed_cluster_gpus_amd.engine.oracle import OracleEngine
         eng = OracleEngine(sc, arrival_inf, arrival_trn, **common)
