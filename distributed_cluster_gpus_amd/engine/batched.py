@@ -301,6 +301,7 @@ class BatchedEngine:
             t["pp_cursor"] = torch.zeros(R, **i32)
             t["pp_size"] = torch.zeros((R, pp_cap), **f64)
             t["pp_done"] = torch.zeros((R, pp_cap), **f64)
+            t["pp_start"] = torch.zeros((R, pp_cap), **f64)
             t["pp_netlat"] = torch.zeros((R, pp_cap), **f32)
             t["pp_jid"] = torch.zeros((R, pp_cap), **i32)
             t["pp_ing"] = torch.zeros((R, pp_cap), **u8_)

@@ -236,6 +236,8 @@ struct EngineDesc {
   int* pp_cursor;                 // [r] next entry to reallocate
   double* pp_size;                // [r][cap] original size
   double* pp_done;                // [r][cap] units completed at preemption
+  double* pp_start;               // [r][cap] ORIGINAL start time (resume
+                                  //   keeps it: latency spans preemptions)
   float* pp_netlat;               // [r][cap]
   int* pp_jid;                    // [r][cap]
   unsigned char* pp_ing;          // [r][cap]
@@ -1095,7 +1097,7 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
                              const float* s0, int a_dc, int a_g,
                              int mdc, int mg, int n_rew,
                              double units_done = 0.0, int pcount = 0,
-                             int has_rl = 1) {
+                             int has_rl = 1, double orig_start = -1.0) {
   const EngineDesc& S = *c.S;
   int64_t base = (int64_t)c.r * S.total_slots;
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
@@ -1111,16 +1113,23 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
     return;
   }
   double T = d_unit_time(n, f, c.lc3(d, jt));
-  // resume semantics: remaining units at the new (n, f)
-  // (reference _resume_preempted_job, :362-387)
-  double units_left = fmax(0.0, size - units_done);
-  double finish = now + units_left * T;
+  // fresh starts: finish = now + size*T (reference _start_job_with_nf);
+  // resumes (orig_start >= 0): the reference's exact arithmetic
+  // units_left / max(1/T, 1e-9) and the ORIGINAL start time preserved
+  // (reference _resume_preempted_job :362-387 — start_time not reset)
+  double finish;
+  if (orig_start >= 0.0) {
+    double units_left = fmax(0.0, size - units_done);
+    finish = now + units_left / fmax(1.0 / T, 1e-9);
+  } else {
+    finish = now + size * T;
+  }
   // copy s0 trace (lane-parallel over obs_dim)
   for (int k = c.lane; k < S.obs_dim; k += SUBW)
     S.slot_s0[(base + cand) * S.obs_dim + k] = s0[k];
   if (c.lane == 0) {
     c.l_fin[cand] = finish;
-    S.s_start[base + cand] = now;
+    S.s_start[base + cand] = orig_start >= 0.0 ? orig_start : now;
     S.s_lastupd[base + cand] = now;
     S.s_size[base + cand] = size;
     S.s_fused[base + cand] = f;
@@ -1196,8 +1205,16 @@ __device__ int rl_elastic_preempt_all(Ctx& c, int d, double now) {
   int64_t base = (int64_t)c.r * S.total_slots;
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
   int count = 0;
-  for (int k = lo; k < hi; ++k) {   // uniform serial walk (rare event)
-    if (S.s_gpus[base + k] == 0 || S.s_jtype[base + k] != 1) continue;
+  // preempt in running_jobs INSERTION order (ascending s_seq), matching the
+  // reference's dict walk; pool order drives the reallocation sequence
+  for (;;) {   // uniform selection loop (rare event)
+    int k = -1, best_seq = INT_MAX;
+    for (int q = lo; q < hi; ++q) {
+      if (S.s_gpus[base + q] == 0 || S.s_jtype[base + q] != 1) continue;
+      int sq = S.s_seq[base + q];
+      if (sq < best_seq) { best_seq = sq; k = q; }
+    }
+    if (k < 0) break;
     if (count >= S.pp_cap) {
       if (c.lane == 0) atomicOr(&S.err[c.r], ERR_SLOT_OVF);
       break;
@@ -1213,6 +1230,7 @@ __device__ int rl_elastic_preempt_all(Ctx& c, int d, double now) {
     if (c.lane == 0) {
       S.pp_size[pb] = size;
       S.pp_done[pb] = done;
+      S.pp_start[pb] = S.s_start[base + k];
       S.pp_netlat[pb] = S.s_netlat[base + k];
       S.pp_jid[pb] = S.s_jid[base + k];
       S.pp_ing[pb] = (unsigned char)S.s_ing[base + k];
@@ -1390,7 +1408,7 @@ __device__ void rl_realloc_inline(Ctx& c, double now) {
                    S.pp_jid[pb], S.pp_ing[pb], n_rl, f, now,
                    &S.pp_s0[pb * S.obs_dim], S.pp_adc[pb], S.pp_ag[pb],
                    mdc, mg, n_rl, S.pp_done[pb], S.pp_pcount[pb],
-                   S.pp_has_rl[pb]);
+                   S.pp_has_rl[pb], S.pp_start[pb]);
     }
   }
   if (c.lane == 0) {
@@ -1529,7 +1547,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
                      S.pp_jid[pb], S.pp_ing[pb], n_rl, f, c.now,
                      &S.pp_s0[pb * S.obs_dim], S.pp_adc[pb], S.pp_ag[pb],
                      mdc, mg, n_rl, S.pp_done[pb], S.pp_pcount[pb],
-                     S.pp_has_rl[pb]);
+                     S.pp_has_rl[pb], S.pp_start[pb]);
       }
       int nxt = cur + 1;
       if (nxt < S.pp_count[c.r]) {
@@ -2196,7 +2214,8 @@ class BatchedSimHip {
       S_.slot_has_rl = reinterpret_cast<unsigned char*>(t_["slot_has_rl"].data_ptr<uint8_t>());
       S_.slot_nrew = reinterpret_cast<unsigned char*>(t_["slot_nrew"].data_ptr<uint8_t>());
       T_PTR(pp_count, int); T_PTR(pp_cursor, int);
-      T_PTR(pp_size, double); T_PTR(pp_done, double); T_PTR(pp_netlat, float);
+      T_PTR(pp_size, double); T_PTR(pp_done, double); T_PTR(pp_start, double);
+      T_PTR(pp_netlat, float);
       T_PTR(pp_jid, int);
       S_.pp_ing = reinterpret_cast<unsigned char*>(t_["pp_ing"].data_ptr<uint8_t>());
       S_.pp_dc = reinterpret_cast<unsigned char*>(t_["pp_dc"].data_ptr<uint8_t>());

@@ -699,7 +699,14 @@ def test_chsac_device_vs_host_serving_consistent():
 
 
 @needs_gpu
-def test_chsac_pinned_policy_parity(tmp_path):
+@pytest.mark.parametrize("inf_rate,trn_rate,elastic,duration", [
+    (2.5, 0.3, False, 150.0),
+    # elastic variant: training-only load long enough for completions, so
+    # preempt-all + deterministic reallocation chains run on BOTH engines
+    (0.0, 0.5, True, 1200.0),
+])
+def test_chsac_pinned_policy_parity(tmp_path, inf_rate, trn_rate, elastic,
+                                    duration):
     """with a FROZEN agent served greedily (deterministic),
     the exact-p99 window enabled, and the oracle's recorded arrival trace,
     one GPU replica reproduces the oracle's chsac trajectory event-for-event
@@ -714,19 +721,20 @@ def test_chsac_pinned_policy_parity(tmp_path):
     from distributed_cluster_gpus_amd.rl.agent import (CHSACAgent,
                                                        CHSACAgentConfig)
     torch.manual_seed(42)
-    duration = 150.0
     agent = CHSACAgent(CHSACAgentConfig(
         obs_dim=49, n_dc=8, n_g_choices=8,
         constraints={"latency_p99": 500.0, "gpu_over": 0.0}, device="cuda"))
     sc = paper_scenario()
-    inf = ArrivalProcess(mode="poisson", rate=2.5)
-    trn = ArrivalProcess(mode="poisson", rate=0.3)
+    inf = ArrivalProcess(mode="poisson" if inf_rate > 0 else "off",
+                         rate=inf_rate)
+    trn = ArrivalProcess(mode="poisson", rate=trn_rate)
     rec = []
     out_o = str(tmp_path / "oracle")
     o_eng = OracleEngine(sc, inf, trn, algo="chsac_af", duration=duration,
                          log_interval=5.0, out_dir=out_o, seed=123,
                          rl_agent=agent, rl_warmup=10**9, rl_device="cuda",
-                         rl_deterministic=True, arrival_recorder=rec)
+                         rl_deterministic=True, arrival_recorder=rec,
+                         elastic_scaling=elastic)
     o_eng.run()
 
     ing_idx = {n: i for i, n in enumerate(sc.ingress_names)}
@@ -750,6 +758,7 @@ def test_chsac_pinned_policy_parity(tmp_path):
                         seed=999, enable_logs=True, rl_agent=agent,
                         rl_warmup=10**9, rl_serve="host",
                         rl_deterministic=True, rl_exact_p99=True,
+                        elastic_scaling=elastic,
                         arrival_trace=(times, sizes))
     eng.run()
 
@@ -770,8 +779,12 @@ def test_chsac_pinned_policy_parity(tmp_path):
     # transition-stream parity: same count, same actions/rewards/costs/masks
     # (costs carry the EXACT sliding-window p99 -> this pins the exact-p99
     # device path against the reference's np.percentile)
+    if elastic:
+        assert (jo["preempt_count"] > 0).any(), \
+            "elastic parity is vacuous: no preemption occurred"
     orep, grep_ = o_eng.replay, eng.replay
-    assert orep.size == grep_.size > 100, (orep.size, grep_.size)
+    assert orep.size == grep_.size > (20 if elastic else 100), \
+        (orep.size, grep_.size)
     n = orep.size
     assert torch.equal(orep.a_dc[:n].cpu(), grep_.a_dc[:n].cpu())
     assert torch.equal(orep.a_g[:n].cpu(), grep_.a_g[:n].cpu())
