@@ -66,6 +66,33 @@ def test_engine_runs_and_conserves():
     # every replica processed a similar order of events (same workload)
     evs = eng.t["ev_count"].cpu().numpy()
     assert evs.min() > 0.5 * evs.mean()
+    assert math.isfinite(st["mean_wait_s"]) and st["mean_wait_s"] >= 0.0
+
+
+@needs_gpu
+def test_queueing_delay_metric_positive_under_congestion():
+    """sum_wait measures real queueing delay: a config that saturates DCs
+    (every job pinned to 8 GPUs, heavy arrivals) must report positive mean
+    wait, and an uncongested one ~zero."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="poisson", rate=30.0)
+    trn = ArrivalProcess(mode="poisson", rate=2.0)
+    eng = BatchedEngine(sc, inf, trn, algo="debug", replicas=16,
+                        duration=120.0, log_interval=5.0, out_dir=None,
+                        seed=11, enable_logs=False, num_fixed_gpus=8,
+                        fixed_freq=0.3, tcap=256)
+    st = eng.run()
+    assert st["mean_wait_s"] > 0.0, "no wait recorded under congestion"
+    inf2 = ArrivalProcess(mode="poisson", rate=0.2)
+    trn2 = ArrivalProcess(mode="off", rate=0.0)
+    eng2 = BatchedEngine(sc, inf2, trn2, algo="default_policy", replicas=16,
+                         duration=120.0, log_interval=5.0, out_dir=None,
+                         seed=11, enable_logs=False)
+    st2 = eng2.run()
+    assert st2["mean_wait_s"] < st["mean_wait_s"]
 
 
 @needs_gpu
