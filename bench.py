@@ -115,30 +115,17 @@ def measure_rl_loop(device, wall_budget_s=8.0, replicas=4096):
                         seed=1, enable_logs=False, rl_warmup=2048,
                         rl_batch=256, rl_train_interval=256,
                         rl_stats_interval=0, events_per_launch=100000)
-    t = eng.t
-
-    def cycle():
-        eng._sim.advance(eng.end_time, eng.events_per_launch)
-        status = torch.stack([t["err"].max(), t["tr_count"][0]]).cpu()
-        assert int(status[0]) == 0, "engine error in rl bench"
-        n_new = eng._rl_ingest(n_tr=int(status[1]))
-        eng._tr_backlog = getattr(eng, "_tr_backlog", 0) + n_new
-        steps = 0
-        if eng.replay.size >= eng._rl_warmup:
-            steps = min(64, eng._tr_backlog // eng._rl_train_interval)
-            eng._tr_backlog -= steps * eng._rl_train_interval
-        eng._rl_train(steps)
-
-    cycle()  # warmup (captures the hipGraph train step lazily)
+    # warm phase: fill replay past warmup, capture the train graph, settle
+    # clocks — then time the production overlapped loop itself
+    eng.run(max_wall_s=3.0)
     torch.cuda.synchronize(device)
-    ev0 = int(t["ev_count"].sum().item())
+    ev0 = int(eng.t["ev_count"].sum().item())
     up0 = eng.rl_updates
     t0 = _t.perf_counter()
-    while _t.perf_counter() - t0 < wall_budget_s:
-        cycle()
+    eng.run(max_wall_s=wall_budget_s)
     torch.cuda.synchronize(device)
     el = _t.perf_counter() - t0
-    return ((int(t["ev_count"].sum().item()) - ev0) / el,
+    return ((int(eng.t["ev_count"].sum().item()) - ev0) / el,
             (eng.rl_updates - up0) / el)
 
 

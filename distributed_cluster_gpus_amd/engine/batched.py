@@ -466,8 +466,19 @@ class BatchedEngine:
         return out
 
     # ---------------- run ----------------
-    def run(self):
-        """Drive advance launches until every replica reaches end_time.
+    def run(self, max_wall_s: Optional[float] = None):
+        """Drive advance launches until every replica reaches end_time (or
+        `max_wall_s` of wall clock passes — benchmarking aid).
+
+        Single-GPU chsac with in-kernel serving runs the OVERLAPPED loop:
+        the advance kernel executes on its own HIP stream while SAC train
+        steps replay on the default stream, so training no longer serializes
+        with simulation; the serving weight buffer refreshes only between
+        launches (the kernel never reads a half-copied buffer).  Training is
+        then paced opportunistically (as many updates as fit under the
+        advance window) — with thousands of replicas the effective
+        samples-trained-per-transition ratio stays ~1 like the
+        interval-paced cadence, but wall-clock decouples.
 
         Data-parallel mode (world > 1 with torch.distributed initialized):
         every rank executes the SAME number of loop iterations and, inside
@@ -480,13 +491,26 @@ class BatchedEngine:
         t = self.t
         from ..parallel.dist import dp_sync_step, is_distributed
         dp = self.is_rl and self.world > 1 and is_distributed()
+        overlap = self.is_rl and not dp and self._serve_device
+        adv_stream = torch.cuda.Stream(device=self.device) if overlap else None
         launches = 0
         self._tr_backlog = 0  # transitions not yet converted into train steps
         import time as _time
         tm = {"advance_s": 0.0, "serve_s": 0.0, "dp_sync_s": 0.0,
-              "train_s": 0.0, "launches": 0}
+              "train_s": 0.0, "overlap_train_steps": 0, "launches": 0}
         self.timing = tm
+        wall0 = _time.perf_counter()
         while True:
+            if overlap:
+                if self._run_cycle_overlapped(adv_stream, tm, _time):
+                    break
+                launches += 1
+                tm["launches"] = launches
+                if launches > 1000000:
+                    raise RuntimeError("batched engine failed to converge")
+                if max_wall_s and _time.perf_counter() - wall0 > max_wall_s:
+                    break
+                continue
             t0 = _time.perf_counter()
             self._sim.advance(self.end_time, self.events_per_launch)
             launches += 1
@@ -550,11 +574,57 @@ class BatchedEngine:
                     break
             if launches > 1000000:
                 raise RuntimeError("batched engine failed to converge")
+            if max_wall_s and _time.perf_counter() - wall0 > max_wall_s:
+                break
         self.meter.count = int(t["ev_count"].sum().item())
         self.meter.stop()
         if self.log_replica >= 0 and self.out_dir is not None:
             self._write_logs()
         return self.stats()
+
+    def _run_cycle_overlapped(self, adv_stream, tm, _time):
+        """One overlapped cycle: advance on adv_stream, SAC updates on the
+        default stream while the kernel runs, then sync / ingest / refresh.
+        Returns True when every replica is done."""
+        t = self.t
+        t0 = _time.perf_counter()
+        # order the advance after last cycle's weight refresh
+        adv_stream.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(adv_stream):
+            self._sim.advance(self.end_time, self.events_per_launch)
+        done_ev = torch.cuda.Event()
+        done_ev.record(adv_stream)
+        # train under the advance window (one graph replay at a time, synced
+        # so host pacing tracks device completion)
+        t1 = _time.perf_counter()
+        can_train = (self.replay.size >=
+                     max(self._rl_warmup, self._rl_batch))
+        trained = 0
+        if can_train:
+            while trained < 256 and not done_ev.query():
+                self._rl_train(1, refresh=False)
+                torch.cuda.current_stream(self.device).synchronize()
+                trained += 1
+            tm["overlap_train_steps"] += trained
+        done_ev.synchronize()
+        tm["train_s"] += _time.perf_counter() - t1
+        torch.cuda.current_stream(self.device).wait_event(done_ev)
+        status = torch.stack([
+            t["err"].max(),
+            t["done"].min(),
+            t["tr_count"][0],
+            t["jl_count"][0],
+        ]).cpu()
+        err = int(status[0])
+        if err != 0:
+            raise RuntimeError(f"batched engine error flags: {err:#x} "
+                               f"(queue/transfer/slot/log overflow)")
+        if int(status[3]) >= int(t["jl_rows"].shape[0]) // 2:
+            self._drain_job_rows()
+        self._rl_ingest(n_tr=int(status[2]))
+        self._refresh_policy_weights()
+        tm["advance_s"] += _time.perf_counter() - t0
+        return int(status[1]) == 1
 
     def _drain_job_rows(self):
         """Chunk-wise drain of the device job-log buffer so unboundedly many
@@ -635,13 +705,16 @@ class BatchedEngine:
         t["tr_count"].zero_()
         return n_tr
 
-    def _rl_train(self, steps: int):
+    def _rl_train(self, steps: int, refresh: bool = True):
         """Run `steps` SAC updates: hipGraph-replayed when captured, eager
         otherwise.  Every `rl_stats_interval`-th update runs eagerly with
         compute_stats=True and logs losses/alpha/lambda at INFO — the
         batched-path equivalent of the reference's per-update INFO logging
         (simulator_paper_multi.py:755,807; sampled here because production
-        updates are graph-replayed and sync-free)."""
+        updates are graph-replayed and sync-free).  refresh=False defers the
+        serving-buffer update to the caller (the overlapped loop refreshes
+        only after the concurrent advance kernel has finished, so the kernel
+        never reads a half-copied weight buffer)."""
         if steps <= 0:
             return
         if self._use_graph and self._graphed is None:
@@ -664,7 +737,7 @@ class BatchedEngine:
                 # sync-free eager SAC step (no stats, tensorized PID)
                 self.rl.train_step(self.replay.sample(self._rl_batch),
                                    compute_stats=False)
-        if self._serve_device:
+        if refresh and self._serve_device:
             self._refresh_policy_weights()
 
     # ---- in-kernel serving support ----
