@@ -370,8 +370,12 @@ class BatchedEngine:
             t["policy_weights"] = torch.zeros(wsize if self._serve_device
                                               else 1, **f32)
             self._rl_hid = H
+            # default yield point: long enough that the overlapped loop can
+            # sustain back-to-back train-graph replays under the advance
+            # window (updates/s stays at the device train rate), short
+            # enough to bound policy staleness to one cycle of transitions
             self._tr_limit = int(rl_tr_limit) if rl_tr_limit is not None else \
-                min(int(tr_cap) // 2, self._rl_train_interval * 32)
+                min(int(tr_cap) // 2, self._rl_train_interval * 256)
         self.t = t
         self.arrival_inf, self.arrival_trn = arrival_inf, arrival_trn
 

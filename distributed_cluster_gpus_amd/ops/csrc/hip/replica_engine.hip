@@ -1407,7 +1407,7 @@ __device__ void rl_realloc_inline(Ctx& c, double now) {
       rl_start_job(c, d_src, 1, S.pp_size[pb], S.pp_netlat[pb],
                    S.pp_jid[pb], S.pp_ing[pb], n_rl, f, now,
                    &S.pp_s0[pb * S.obs_dim], S.pp_adc[pb], S.pp_ag[pb],
-                   mdc, mg, n_rl, S.pp_done[pb], S.pp_pcount[pb],
+                   mdc, mg, S.pp_nrew[pb], S.pp_done[pb], S.pp_pcount[pb],
                    S.pp_has_rl[pb], S.pp_start[pb]);
     }
   }
@@ -1546,7 +1546,7 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         rl_start_job(c, d_src, 1, S.pp_size[pb], S.pp_netlat[pb],
                      S.pp_jid[pb], S.pp_ing[pb], n_rl, f, c.now,
                      &S.pp_s0[pb * S.obs_dim], S.pp_adc[pb], S.pp_ag[pb],
-                     mdc, mg, n_rl, S.pp_done[pb], S.pp_pcount[pb],
+                     mdc, mg, S.pp_nrew[pb], S.pp_done[pb], S.pp_pcount[pb],
                      S.pp_has_rl[pb], S.pp_start[pb]);
       }
       int nxt = cur + 1;

@@ -762,6 +762,15 @@ def test_chsac_pinned_policy_parity(tmp_path, inf_rate, trn_rate, elastic,
                          rl_agent=agent, rl_warmup=10**9, rl_device="cuda",
                          rl_deterministic=True, arrival_recorder=rec,
                          elastic_scaling=elastic)
+    # capture the oracle's reward-n per transition for divergence triage
+    o_paths = []
+    _orig_fin = o_eng._rl_on_finish
+
+    def _fin(dc, job, g, f_used, rl_metrics):
+        o_paths.append((int(job.jid), int(job.rl_action["n"]), int(g),
+                        float(o_eng.now)))
+        return _orig_fin(dc, job, g, f_used, rl_metrics)
+    o_eng._rl_on_finish = _fin
     o_eng.run()
 
     ing_idx = {n: i for i, n in enumerate(sc.ingress_names)}
@@ -813,6 +822,19 @@ def test_chsac_pinned_policy_parity(tmp_path, inf_rate, trn_rate, elastic,
     assert orep.size == grep_.size > (20 if elastic else 100), \
         (orep.size, grep_.size)
     n = orep.size
+    # triage dump: full transition streams + oracle path info (merged back
+    # through gpurun_out for offline analysis if an assert below trips)
+    dbg_dir = os.path.join(REPO, "gpurun_out")
+    if os.path.isdir(dbg_dir):
+        np.savez(os.path.join(dbg_dir, f"parity_dbg_el{int(elastic)}.npz"),
+                 r_o=orep.r[:n].cpu().numpy(), r_g=grep_.r[:n].cpu().numpy(),
+                 s0_o=orep.s[:n].cpu().numpy(), s0_g=grep_.s[:n].cpu().numpy(),
+                 sn_o=orep.s_next[:n].cpu().numpy(),
+                 sn_g=grep_.s_next[:n].cpu().numpy(),
+                 co=orep.costs[:n].cpu().numpy(),
+                 cg=grep_.costs[:n].cpu().numpy(),
+                 adc=orep.a_dc[:n].cpu().numpy(), ag=orep.a_g[:n].cpu().numpy(),
+                 opaths=np.asarray(o_paths, np.float64))
     assert torch.equal(orep.a_dc[:n].cpu(), grep_.a_dc[:n].cpu())
     assert torch.equal(orep.a_g[:n].cpu(), grep_.a_g[:n].cpu())
     assert torch.allclose(orep.r[:n].cpu(), grep_.r[:n].cpu(),
