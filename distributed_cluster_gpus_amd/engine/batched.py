@@ -496,7 +496,13 @@ class BatchedEngine:
         from ..parallel.dist import dp_sync_step, is_distributed
         dp = self.is_rl and self.world > 1 and is_distributed()
         overlap = self.is_rl and not dp and self._serve_device
-        adv_stream = torch.cuda.Stream(device=self.device) if overlap else None
+        adv_stream = train_stream = None
+        if overlap:
+            adv_stream = torch.cuda.Stream(device=self.device)
+            # training gets a HIGH-priority stream so its small GEMM kernels
+            # co-schedule into CU slots alongside the long-running advance
+            # blocks instead of starving behind them
+            train_stream = torch.cuda.Stream(device=self.device, priority=-1)
         launches = 0
         self._tr_backlog = 0  # transitions not yet converted into train steps
         import time as _time
@@ -506,7 +512,8 @@ class BatchedEngine:
         wall0 = _time.perf_counter()
         while True:
             if overlap:
-                if self._run_cycle_overlapped(adv_stream, tm, _time):
+                if self._run_cycle_overlapped(adv_stream, train_stream, tm,
+                                              _time):
                     break
                 launches += 1
                 tm["launches"] = launches
@@ -586,13 +593,13 @@ class BatchedEngine:
             self._write_logs()
         return self.stats()
 
-    def _run_cycle_overlapped(self, adv_stream, tm, _time):
+    def _run_cycle_overlapped(self, adv_stream, train_stream, tm, _time):
         """One overlapped cycle: advance on adv_stream, SAC updates on the
-        default stream while the kernel runs, then sync / ingest / refresh.
-        Returns True when every replica is done."""
+        high-priority train stream while the kernel runs, then sync /
+        ingest / refresh.  Returns True when every replica is done."""
         t = self.t
         t0 = _time.perf_counter()
-        # order the advance after last cycle's weight refresh
+        # order the advance after last cycle's weight refresh / ingest
         adv_stream.wait_stream(torch.cuda.current_stream(self.device))
         with torch.cuda.stream(adv_stream):
             self._sim.advance(self.end_time, self.events_per_launch)
@@ -605,14 +612,19 @@ class BatchedEngine:
                      max(self._rl_warmup, self._rl_batch))
         trained = 0
         if can_train:
+            train_stream.wait_stream(torch.cuda.current_stream(self.device))
             while trained < 256 and not done_ev.query():
-                self._rl_train(1, refresh=False)
-                torch.cuda.current_stream(self.device).synchronize()
+                with torch.cuda.stream(train_stream):
+                    self._rl_train(1, refresh=False)
+                train_stream.synchronize()
                 trained += 1
             tm["overlap_train_steps"] += trained
         done_ev.synchronize()
         tm["train_s"] += _time.perf_counter() - t1
-        torch.cuda.current_stream(self.device).wait_event(done_ev)
+        cur = torch.cuda.current_stream(self.device)
+        cur.wait_event(done_ev)
+        if can_train:
+            cur.wait_stream(train_stream)
         status = torch.stack([
             t["err"].max(),
             t["done"].min(),
@@ -727,7 +739,8 @@ class BatchedEngine:
         for _ in range(steps):
             self.rl_updates += 1
             want_stats = (self._rl_stats_interval > 0 and self.logger is not None
-                          and self.rl_updates % self._rl_stats_interval == 0)
+                          and (self.rl_updates == 1 or
+                               self.rl_updates % self._rl_stats_interval == 0))
             if want_stats:
                 stats = self.rl.train_step(self.replay.sample(self._rl_batch),
                                            compute_stats=True)

@@ -390,6 +390,11 @@ def test_chsac_batched_via_cli(tmp_path):
                        capture_output=True, text=True, timeout=600, cwd=REPO)
     assert r.returncode == 0, r.stderr[-2000:]
     assert os.path.exists(os.path.join(out, "job_log.csv"))
+    # RL observability: SAC losses/alpha reach project.log at INFO on the
+    # batched production path (round-1 VERDICT item 7)
+    with open(os.path.join(out, "project.log")) as fh:
+        log_text = fh.read()
+    assert "loss_critic" in log_text, "no SAC stats in project.log"
 
 
 @needs_gpu
@@ -751,6 +756,15 @@ def test_chsac_pinned_policy_parity(tmp_path, inf_rate, trn_rate, elastic,
     agent = CHSACAgent(CHSACAgentConfig(
         obs_dim=49, n_dc=8, n_g_choices=8,
         constraints={"latency_p99": 500.0, "gpu_over": 0.0}, device="cuda"))
+    # pin the greedy policy to (dc=2 — the 256-GPU DC, g=7 -> n=8): a frozen
+    # random actor can wedge every job into a 16-GPU DC and starve the run;
+    # this keeps churn high AND exercises free-GPU clamping on drains
+    with torch.no_grad():
+        for head, idx in ((agent.actor.head_dc[2], 2),
+                          (agent.actor.head_g[2], 7)):
+            head.weight.zero_()
+            head.bias.zero_()
+            head.bias[idx] = 10.0
     sc = paper_scenario()
     inf = ArrivalProcess(mode="poisson" if inf_rate > 0 else "off",
                          rate=inf_rate)
@@ -819,9 +833,7 @@ def test_chsac_pinned_policy_parity(tmp_path, inf_rate, trn_rate, elastic,
         assert (jo["preempt_count"] > 0).any(), \
             "elastic parity is vacuous: no preemption occurred"
     orep, grep_ = o_eng.replay, eng.replay
-    assert orep.size == grep_.size > (20 if elastic else 100), \
-        (orep.size, grep_.size)
-    n = orep.size
+    n = min(orep.size, grep_.size)
     # triage dump: full transition streams + oracle path info (merged back
     # through gpurun_out for offline analysis if an assert below trips)
     dbg_dir = os.path.join(REPO, "gpurun_out")
@@ -835,6 +847,8 @@ def test_chsac_pinned_policy_parity(tmp_path, inf_rate, trn_rate, elastic,
                  cg=grep_.costs[:n].cpu().numpy(),
                  adc=orep.a_dc[:n].cpu().numpy(), ag=orep.a_g[:n].cpu().numpy(),
                  opaths=np.asarray(o_paths, np.float64))
+    assert orep.size == grep_.size > (20 if elastic else 100), \
+        (orep.size, grep_.size)
     assert torch.equal(orep.a_dc[:n].cpu(), grep_.a_dc[:n].cpu())
     assert torch.equal(orep.a_g[:n].cpu(), grep_.a_g[:n].cpu())
     assert torch.allclose(orep.r[:n].cpu(), grep_.r[:n].cpu(),
