@@ -2251,6 +2251,50 @@ class BatchedSimHip {
     }
   }
 
+  // Create a CU-masked stream for the advance kernel, excluding the LAST
+  // `n_reserved` CUs.  The advance kernel's blocks are persistent for a
+  // whole cycle (each wave loops until its replica yields), so without a
+  // mask they occupy every wave slot on the device and concurrent work
+  // (the overlapped SAC train kernels) starves behind them — stream
+  // priorities cannot help because no slot ever frees.  Masking carves a
+  // small CU island the advance never touches; kernels on ordinary streams
+  // land there immediately.
+  void enable_masked_stream(int n_reserved) {
+    if (masked_stream_) return;
+    hipDeviceProp_t prop;
+    int dev;
+    (void)hipGetDevice(&dev);
+    (void)hipGetDeviceProperties(&prop, dev);
+    int n_cu = prop.multiProcessorCount;
+    int words = (n_cu + 31) / 32;
+    std::vector<uint32_t> mask(words, 0u);
+    int usable = n_cu - n_reserved;
+    if (usable < 1) usable = 1;
+    for (int i = 0; i < usable; ++i) mask[i / 32] |= (1u << (i % 32));
+    hipError_t e = hipExtStreamCreateWithCUMask(&masked_stream_,
+                                                (uint32_t)words, mask.data());
+    if (e != hipSuccess) {
+      masked_stream_ = nullptr;  // fall back to the torch stream
+      throw std::runtime_error(std::string("CU-mask stream: ") +
+                               hipGetErrorString(e));
+    }
+  }
+
+  bool advance_done() {
+    hipStream_t s = masked_stream_ ? masked_stream_
+                                   : (hipStream_t)at::hip::getCurrentHIPStream();
+    return hipStreamQuery(s) == hipSuccess;
+  }
+
+  void advance_sync() {
+    hipStream_t s = masked_stream_ ? masked_stream_
+                                   : (hipStream_t)at::hip::getCurrentHIPStream();
+    hipError_t e = hipStreamSynchronize(s);
+    if (e != hipSuccess)
+      throw std::runtime_error(std::string("advance_sync: ") +
+                               hipGetErrorString(e));
+  }
+
   // launch one advance chunk; returns immediately (stream-async)
   void advance(double t_target, int64_t max_ev) {
     int rpb = REPLICAS_PER_BLOCK;
@@ -2270,7 +2314,8 @@ class BatchedSimHip {
       throw std::runtime_error(
           "scenario too large for the LDS-mirrored engine (total_slots + "
           "tcap exceed the 64 KiB dynamic-LDS budget per block)");
-    hipStream_t stream = at::hip::getCurrentHIPStream();
+    hipStream_t stream = masked_stream_
+        ? masked_stream_ : (hipStream_t)at::hip::getCurrentHIPStream();
     switch (S_.algo) {
       case A_DEFAULT:
         hipLaunchKernelGGL(advance_kernel<A_DEFAULT>, grid, block, shmem, stream, S_, t_target, max_ev); break;
@@ -2302,6 +2347,7 @@ class BatchedSimHip {
  private:
   EngineDesc S_;
   std::unordered_map<std::string, torch::Tensor> t_;
+  hipStream_t masked_stream_ = nullptr;
 };
 
 #undef T_PTR
@@ -2316,7 +2362,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   py::class_<dcg::BatchedSimHip>(m, "BatchedSimHip", py::module_local())
       .def(py::init<py::dict, py::dict>())
       .def("advance", &dcg::BatchedSimHip::advance,
-           py::arg("t_target"), py::arg("max_events"));
+           py::arg("t_target"), py::arg("max_events"))
+      .def("enable_masked_stream", &dcg::BatchedSimHip::enable_masked_stream,
+           py::arg("n_reserved_cus"))
+      .def("advance_done", &dcg::BatchedSimHip::advance_done)
+      .def("advance_sync", &dcg::BatchedSimHip::advance_sync);
   m.def("rl_forward_debug", &dcg::rl_forward_debug,
         "batched actor forward via the in-kernel serving math "
         "(pw, obs[B,D], hid, n_dc, n_g) -> (logits_dc, logits_g)",
