@@ -733,6 +733,11 @@ def test_chsac_device_vs_host_serving_consistent():
 @needs_gpu
 @pytest.mark.parametrize("inf_rate,trn_rate,elastic,duration", [
     (2.5, 0.3, False, 150.0),
+    # saturating variant: heavy arrivals partially fill the pinned DC, so
+    # admissions and drains get CLAMPED below the pinned n=8 — the regime
+    # where reward-n bookkeeping (unclamped at arrival, clamped at drain)
+    # can silently diverge
+    (8.0, 0.3, False, 120.0),
     # elastic variant: training-only load long enough for completions, so
     # preempt-all + deterministic reallocation chains run on BOTH engines
     (0.0, 0.5, True, 1200.0),
@@ -838,7 +843,8 @@ def test_chsac_pinned_policy_parity(tmp_path, inf_rate, trn_rate, elastic,
     # through gpurun_out for offline analysis if an assert below trips)
     dbg_dir = os.path.join(REPO, "gpurun_out")
     if os.path.isdir(dbg_dir):
-        np.savez(os.path.join(dbg_dir, f"parity_dbg_el{int(elastic)}.npz"),
+        np.savez(os.path.join(dbg_dir,
+                              f"parity_dbg_el{int(elastic)}_r{inf_rate}.npz"),
                  r_o=orep.r[:n].cpu().numpy(), r_g=grep_.r[:n].cpu().numpy(),
                  s0_o=orep.s[:n].cpu().numpy(), s0_g=grep_.s[:n].cpu().numpy(),
                  sn_o=orep.s_next[:n].cpu().numpy(),
