@@ -304,6 +304,11 @@ constexpr int THREADS_PER_BLOCK = 128;
 constexpr int REPLICAS_PER_BLOCK = THREADS_PER_BLOCK / DCG_SUBWAVE;
 struct Hot {
   double arr_next[MAX_STREAMS];
+  // per-replica scalars updated every arrival/finish — LDS-resident so the
+  // hot path never does global read-modify-writes for them
+  double sum_lat, sum_lat_inf, sum_wait;
+  long long jobs_done, jobs_done_inf;
+  int jid_ctr, seq_ctr;
   double dc_minf[MAX_DC];
   double p_active[MAX_DC];
   double sum_tpt[MAX_DC];
@@ -407,8 +412,7 @@ __device__ void start_job(Ctx& c, int d, int jt, double size, float netlat,
     S.s_ing[base + cand] = (char)ing;
     S.s_done[base + cand] = 0.0;
     S.s_pcount[base + cand] = 0;
-    int sq = S.seq_ctr[c.r] + 1;
-    S.seq_ctr[c.r] = sq;
+    int sq = ++c.hs->seq_ctr;
     S.s_seq[base + cand] = sq;
     c.hs->busy[d] += n;
     c.hs->n_running[d] += 1;
@@ -619,7 +623,7 @@ __device__ void drain_queues(Ctx& c, int d, double now) {
     }
     n = max(1, min(n, c.free_gpus(d)));
     start_job(c, d, jt, size, netlat, jid, ing, n, f, now);
-    if (c.lane == 0) S.sum_wait[c.r] += fmax(0.0, now - enq);
+    if (c.lane == 0) c.hs->sum_wait += fmax(0.0, now - enq);
   }
 }
 
@@ -1140,8 +1144,7 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
     S.s_ing[base + cand] = (char)ing;
     S.s_done[base + cand] = units_done;
     S.s_pcount[base + cand] = (unsigned char)pcount;
-    int sq = S.seq_ctr[c.r] + 1;
-    S.seq_ctr[c.r] = sq;
+    int sq = ++c.hs->seq_ctr;
     S.s_seq[base + cand] = sq;
     S.slot_adc[base + cand] = (unsigned char)a_dc;
     S.slot_ag[base + cand] = (unsigned char)a_g;
@@ -1359,7 +1362,7 @@ __device__ void rl_do_drain_action(Ctx& c, double now, int src_d,
     double f = rl_energy_freq(c, d_tgt, jt, n_sel);
     rl_start_job(c, d_tgt, jt, size, netlat, jid, ing, n_sel, f, now,
                  s0, a_dc, a_g, mdc, mg, n_sel);
-    if (c.lane == 0) S.sum_wait[c.r] += fmax(0.0, now - enq);
+    if (c.lane == 0) c.hs->sum_wait += fmax(0.0, now - enq);
   }
 }
 
@@ -1507,7 +1510,16 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       h->q_len[k] = S.q_len[c.r * nd * 2 + k];
       h->q_head[k] = S.q_head[c.r * nd * 2 + k];
     }
-    if (lane == 0) h->next_log = S.next_log[c.r];
+    if (lane == 0) {
+      h->next_log = S.next_log[c.r];
+      h->sum_lat = S.sum_lat[c.r];
+      h->sum_lat_inf = S.sum_lat_inf[c.r];
+      h->sum_wait = S.sum_wait[c.r];
+      h->jobs_done = S.jobs_done[c.r];
+      h->jobs_done_inf = S.jobs_done_inf[c.r];
+      h->jid_ctr = S.jid_ctr[c.r];
+      h->seq_ctr = S.seq_ctr[c.r];
+    }
   }
   store_fence();  // global loads (vmcnt) AND the LDS writes (lgkmcnt)
 
@@ -1648,8 +1660,8 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       // ===== arrival at ingress stream idx =====
       int ing = idx >> 1;
       int jt = idx & 1;
-      int jid = S.jid_ctr[c.r] + 1;
-      if (lane == 0) S.jid_ctr[c.r] = jid;
+      int jid = c.hs->jid_ctr + 1;
+      if (lane == 0) c.hs->jid_ctr = jid;
       double size;
       int trace_d = -1;
       if (S.trace_mode) {
@@ -1881,11 +1893,11 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         // remainder job-units: window = finish mod log_interval (quirk)
         c.hs->acc_unit[d] += (1.0 / T) * fmod(t_min, S.log_interval);
         // metrics
-        S.jobs_done[c.r] += 1;
-        S.sum_lat[c.r] += t_min - start;
+        c.hs->jobs_done += 1;
+        c.hs->sum_lat += t_min - start;
         if (jt == 0) {
-          S.jobs_done_inf[c.r] += 1;
-          S.sum_lat_inf[c.r] += t_min - start;
+          c.hs->jobs_done_inf += 1;
+          c.hs->sum_lat_inf += t_min - start;
         }
         if (ALGO == A_BANDIT) {
           // reward = -E_pred (energy per unit at the used f); match the arm
@@ -2028,6 +2040,13 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     S.now[c.r] = c.now;
     S.rng_ctr[c.r] = c.rng.ctr;
     S.ev_count[c.r] += n_events;
+    S.sum_lat[c.r] = c.hs->sum_lat;
+    S.sum_lat_inf[c.r] = c.hs->sum_lat_inf;
+    S.sum_wait[c.r] = c.hs->sum_wait;
+    S.jobs_done[c.r] = c.hs->jobs_done;
+    S.jobs_done_inf[c.r] = c.hs->jobs_done_inf;
+    S.jid_ctr[c.r] = c.hs->jid_ctr;
+    S.seq_ctr[c.r] = c.hs->seq_ctr;
   }
 }
 

@@ -20,11 +20,15 @@ from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
 out_dir = sys.argv[1] if len(sys.argv) > 1 else "/tmp/week_full"
 sc = paper_scenario()
 inf = ArrivalProcess(mode="sinusoid", rate=6.0, amp=0.6, period=300.0)
-trn = ArrivalProcess(mode="poisson", rate=0.3)
+# canonical week matrix training rate (run.sh: trn-rate 0.02).  The cluster
+# cannot absorb more: training jobs run for hours, so higher rates grow
+# q_train without bound (reference behaviour too — its Python list just
+# never overflows); qcap is sized for the week's training backlog.
+trn = ArrivalProcess(mode="poisson", rate=0.02)
 eng = BatchedEngine(sc, inf, trn, algo="default_policy", replicas=8,
                     duration=604800.0, log_interval=20.0,
                     out_dir=out_dir, seed=123, enable_logs=True,
-                    events_per_launch=500000)
+                    qcap=262144, events_per_launch=500000)
 t0 = time.perf_counter()
 st = eng.run()
 gpu_wall = time.perf_counter() - t0
