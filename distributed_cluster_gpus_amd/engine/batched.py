@@ -505,8 +505,12 @@ class BatchedEngine:
             # there, and the train stream's kernels land immediately.
             try:
                 self._sim.enable_masked_stream(self._reserve_cus)
-            except RuntimeError:
-                pass  # masked stream unavailable: overlap still works, slower
+            except RuntimeError as e:
+                # masked stream unavailable: overlap still works, slower
+                import warnings
+                warnings.warn(f"CU-masked stream unavailable ({e}); "
+                              "train kernels may contend with the advance "
+                              "kernel")
         launches = 0
         self._tr_backlog = 0  # transitions not yet converted into train steps
         import time as _time

@@ -2299,6 +2299,8 @@ class BatchedSimHip {
     }
   }
 
+  bool masked_active() const { return masked_stream_ != nullptr; }
+
   bool advance_done() {
     hipStream_t s = masked_stream_ ? masked_stream_
                                    : (hipStream_t)at::hip::getCurrentHIPStream();
@@ -2384,6 +2386,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("t_target"), py::arg("max_events"))
       .def("enable_masked_stream", &dcg::BatchedSimHip::enable_masked_stream,
            py::arg("n_reserved_cus"))
+      .def("masked_active", &dcg::BatchedSimHip::masked_active)
       .def("advance_done", &dcg::BatchedSimHip::advance_done)
       .def("advance_sync", &dcg::BatchedSimHip::advance_sync);
   m.def("rl_forward_debug", &dcg::rl_forward_debug,
