@@ -37,7 +37,7 @@ t1 = time.perf_counter()
 n_job = sum(1 for _ in open(os.path.join(out_dir, "job_log.csv"))) - 1
 n_cl = sum(1 for _ in open(os.path.join(out_dir, "cluster_log.csv"))) - 1
 print(json.dumps({
-    "sim_days": 7, "arrivals": "sinusoid 6/s + poisson 0.3/s (FULL load)",
+    "sim_days": 7, "arrivals": "sinusoid 6/s amp 0.6 (FULL inference load) + poisson trn 0.02/s",
     "events": st["events"], "wall_s": round(gpu_wall, 2),
     "events_per_sec": round(st["events"] / gpu_wall),
     "jobs_log_replica": int(eng.t["jobs_done"][0].item()),
