@@ -511,6 +511,13 @@ class BatchedEngine:
                 warnings.warn(f"CU-masked stream unavailable ({e}); "
                               "train kernels may contend with the advance "
                               "kernel")
+            # train on a torch-created (NON-BLOCKING) stream: the masked
+            # stream is a blocking stream, and the legacy NULL stream
+            # implicitly synchronizes with all blocking streams — training
+            # on the default stream would serialize behind every advance
+            # launch (measured: exactly floor+1 steps/cycle at every island
+            # size).  Two non-NULL streams have no implicit ordering.
+            self._train_stream = torch.cuda.Stream(device=self.device)
         launches = 0
         self._tr_backlog = 0  # transitions not yet converted into train steps
         import time as _time
@@ -619,11 +626,19 @@ class BatchedEngine:
                      max(self._rl_warmup, self._rl_batch))
         trained = 0
         if can_train:
+            ts = self._train_stream
+            ts.wait_stream(torch.cuda.current_stream(self.device))
             while trained < 1024 and not self._sim.advance_done():
-                self._rl_train(1, refresh=False)
-                torch.cuda.current_stream(self.device).synchronize()
+                with torch.cuda.stream(ts):
+                    self._rl_train(1, refresh=False)
+                ts.synchronize()
                 trained += 1
         self._sim.advance_sync()
+        if can_train:
+            # later default-stream ops (floor steps, refresh) must see the
+            # side-stream parameter updates
+            torch.cuda.current_stream(self.device).wait_stream(
+                self._train_stream)
         tm["train_s"] += _time.perf_counter() - t1
         status = torch.stack([
             t["err"].max(),
