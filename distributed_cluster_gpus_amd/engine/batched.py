@@ -522,7 +522,9 @@ class BatchedEngine:
         self._tr_backlog = 0  # transitions not yet converted into train steps
         import time as _time
         tm = {"advance_s": 0.0, "serve_s": 0.0, "dp_sync_s": 0.0,
-              "train_s": 0.0, "overlap_train_steps": 0, "launches": 0}
+              "train_s": 0.0, "overlap_train_steps": 0, "launches": 0,
+              "sync0_s": 0.0, "kernel_s": 0.0, "status_s": 0.0,
+              "ingest_s": 0.0, "floor_s": 0.0, "refresh_s": 0.0}
         self.timing = tm
         wall0 = _time.perf_counter()
         while True:
@@ -618,6 +620,8 @@ class BatchedEngine:
         t0 = _time.perf_counter()
         # the kernel must see last cycle's weight refresh / tr_count reset
         torch.cuda.current_stream(self.device).synchronize()
+        tm["sync0_s"] += _time.perf_counter() - t0
+        tk = _time.perf_counter()
         self._sim.advance(self.end_time, self.events_per_launch)
         # train under the advance window (one graph replay at a time, synced
         # so host pacing tracks device completion)
@@ -634,12 +638,14 @@ class BatchedEngine:
                 ts.synchronize()
                 trained += 1
         self._sim.advance_sync()
+        tm["kernel_s"] += _time.perf_counter() - tk
         if can_train:
             # later default-stream ops (floor steps, refresh) must see the
             # side-stream parameter updates
             torch.cuda.current_stream(self.device).wait_stream(
                 self._train_stream)
         tm["train_s"] += _time.perf_counter() - t1
+        t2 = _time.perf_counter()
         status = torch.stack([
             t["err"].max(),
             t["done"].min(),
@@ -650,21 +656,28 @@ class BatchedEngine:
         if err != 0:
             raise RuntimeError(f"batched engine error flags: {err:#x} "
                                f"(queue/transfer/slot/log overflow)")
+        tm["status_s"] += _time.perf_counter() - t2
         if int(status[3]) >= int(t["jl_rows"].shape[0]) // 2:
             self._drain_job_rows()
+        t3 = _time.perf_counter()
         n_new = self._rl_ingest(n_tr=int(status[2]))
+        tm["ingest_s"] += _time.perf_counter() - t3
         # interval-paced serial floor: opportunistic steps count against the
         # backlog; a bounded catch-up covers cycles whose window was missed
         self._tr_backlog = max(0, self._tr_backlog + n_new -
                                trained * self._rl_train_interval)
+        t4 = _time.perf_counter()
         if self.replay.size >= max(self._rl_warmup, self._rl_batch):
             owed = min(4, self._tr_backlog // self._rl_train_interval)
             if owed > 0:
                 self._rl_train(owed, refresh=False)
                 self._tr_backlog -= owed * self._rl_train_interval
                 trained += owed
+        tm["floor_s"] += _time.perf_counter() - t4
         tm["overlap_train_steps"] += trained
+        t5 = _time.perf_counter()
         self._refresh_policy_weights()
+        tm["refresh_s"] += _time.perf_counter() - t5
         tm["advance_s"] += _time.perf_counter() - t0
         return int(status[1]) == 1
 

@@ -36,8 +36,8 @@ for rc in [int(x) for x in (sys.argv[1:] or ["16", "48", "96", "144"])]:
         "reserve_cus": rc,
         "events_per_sec": round((int(eng.t["ev_count"].sum().item()) - ev0) / el),
         "updates_per_sec": round((eng.rl_updates - up0) / el, 1),
-        "overlap_steps": eng.timing.get("overlap_train_steps"),
-        "launches": eng.timing.get("launches"),
+        **{k: (round(v, 3) if isinstance(v, float) else v)
+           for k, v in eng.timing.items()},
     }), flush=True)
     del eng
     torch.cuda.empty_cache()
