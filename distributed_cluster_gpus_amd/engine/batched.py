@@ -65,6 +65,7 @@ class BatchedEngine:
                  rl_serve: str = "device", rl_deterministic: bool = False,
                  rl_exact_p99: bool = False, fp32_coeff_eval: bool = False,
                  rl_tr_limit: Optional[int] = None, rl_reserve_cus: int = 16,
+                 rl_target_updates_per_s: float = 220.0,
                  tr_cap: int = 262144, arrival_trace=None,
                  subwave: int = 64, **_unused):
         if algo not in ALGOS:
@@ -262,6 +263,12 @@ class BatchedEngine:
         # (used for parity testing and injected non-standard agents)
         self._serve_device = self.is_rl and rl_serve == "device"
         self._reserve_cus = int(rl_reserve_cus)
+        # overlapped-loop update-rate target: after each advance window the
+        # cycle trains serially until the measured SAC update rate meets
+        # this (<=0 disables the controller: throughput mode, training is
+        # purely opportunistic).  220/s is the standalone hipGraph rate, so
+        # the default keeps updates/s at parity with a dedicated trainer.
+        self._rl_target_ups = float(rl_target_updates_per_s)
         if self.is_rl:
             t["req_flag"] = torch.zeros(R, **i32)
             t["req_obs"] = torch.zeros((R, obs_dim), **f32)
@@ -520,6 +527,7 @@ class BatchedEngine:
             self._train_stream = torch.cuda.Stream(device=self.device)
         launches = 0
         self._tr_backlog = 0  # transitions not yet converted into train steps
+        self._ups_t0 = None   # update-rate controller epoch (first trainable)
         import time as _time
         tm = {"advance_s": 0.0, "serve_s": 0.0, "dp_sync_s": 0.0,
               "train_s": 0.0, "overlap_train_steps": 0, "launches": 0,
@@ -660,19 +668,26 @@ class BatchedEngine:
         if int(status[3]) >= int(t["jl_rows"].shape[0]) // 2:
             self._drain_job_rows()
         t3 = _time.perf_counter()
-        n_new = self._rl_ingest(n_tr=int(status[2]))
+        self._rl_ingest(n_tr=int(status[2]))
         tm["ingest_s"] += _time.perf_counter() - t3
-        # interval-paced serial floor: opportunistic steps count against the
-        # backlog; a bounded catch-up covers cycles whose window was missed
-        self._tr_backlog = max(0, self._tr_backlog + n_new -
-                               trained * self._rl_train_interval)
+        # update-rate controller: train serially until the measured update
+        # rate meets the target (the advance kernel's oversubscribed grid
+        # head-of-line-blocks concurrent dispatch, so opportunistic overlap
+        # alone cannot sustain the standalone train rate — measured)
         t4 = _time.perf_counter()
         if self.replay.size >= max(self._rl_warmup, self._rl_batch):
-            owed = min(4, self._tr_backlog // self._rl_train_interval)
-            if owed > 0:
-                self._rl_train(owed, refresh=False)
-                self._tr_backlog -= owed * self._rl_train_interval
-                trained += owed
+            if self._ups_t0 is None:
+                self._ups_t0 = _time.perf_counter()
+                self._ups_base = self.rl_updates
+            tgt = self._rl_target_ups
+            extra = 0
+            while tgt > 0 and extra < 48:
+                el = max(1e-6, _time.perf_counter() - self._ups_t0)
+                if (self.rl_updates - self._ups_base) >= tgt * el:
+                    break
+                self._rl_train(1, refresh=False)
+                extra += 1
+            trained += extra
         tm["floor_s"] += _time.perf_counter() - t4
         tm["overlap_train_steps"] += trained
         t5 = _time.perf_counter()

@@ -14,7 +14,7 @@ from distributed_cluster_gpus_amd.configs.paper import paper_scenario
 from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
 from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
 
-for rc in [int(x) for x in (sys.argv[1:] or ["16", "48", "96", "144"])]:
+for tgt in [float(x) for x in (sys.argv[1:] or ["220", "150", "0"])]:
     sc = paper_scenario()
     inf = ArrivalProcess(mode="sinusoid", rate=6.0, amp=0.6, period=300.0)
     trn = ArrivalProcess(mode="poisson", rate=0.3)
@@ -23,7 +23,7 @@ for rc in [int(x) for x in (sys.argv[1:] or ["16", "48", "96", "144"])]:
                         seed=1, enable_logs=False, rl_warmup=2048,
                         rl_batch=256, rl_train_interval=256,
                         rl_stats_interval=0, events_per_launch=100000,
-                        rl_reserve_cus=rc)
+                        rl_target_updates_per_s=tgt)
     eng.run(max_wall_s=3.0)
     torch.cuda.synchronize()
     ev0 = int(eng.t["ev_count"].sum().item())
@@ -33,7 +33,7 @@ for rc in [int(x) for x in (sys.argv[1:] or ["16", "48", "96", "144"])]:
     torch.cuda.synchronize()
     el = time.perf_counter() - t0
     print(json.dumps({
-        "reserve_cus": rc,
+        "target_ups": tgt,
         "events_per_sec": round((int(eng.t["ev_count"].sum().item()) - ev0) / el),
         "updates_per_sec": round((eng.rl_updates - up0) / el, 1),
         **{k: (round(v, 3) if isinstance(v, float) else v)
