@@ -65,7 +65,7 @@ class BatchedEngine:
                  rl_serve: str = "device", rl_deterministic: bool = False,
                  rl_exact_p99: bool = False, fp32_coeff_eval: bool = False,
                  rl_tr_limit: Optional[int] = None, rl_reserve_cus: int = 16,
-                 rl_target_updates_per_s: float = 220.0,
+                 rl_target_updates_per_s: float = 200.0,
                  tr_cap: int = 262144, arrival_trace=None,
                  subwave: int = 64, **_unused):
         if algo not in ALGOS:
@@ -266,8 +266,9 @@ class BatchedEngine:
         # overlapped-loop update-rate target: after each advance window the
         # cycle trains serially until the measured SAC update rate meets
         # this (<=0 disables the controller: throughput mode, training is
-        # purely opportunistic).  220/s is the standalone hipGraph rate, so
-        # the default keeps updates/s at parity with a dedicated trainer.
+        # purely opportunistic).  The standalone hipGraph rate is ~227/s;
+        # 200 keeps ~1.2M events/s alongside (measured tradeoff curve in
+        # profiles/README.md — 150 -> 2.1M ev/s, 0 -> 5.4M ev/s).
         self._rl_target_ups = float(rl_target_updates_per_s)
         if self.is_rl:
             t["req_flag"] = torch.zeros(R, **i32)
@@ -681,7 +682,7 @@ class BatchedEngine:
                 self._ups_base = self.rl_updates
             tgt = self._rl_target_ups
             extra = 0
-            while tgt > 0 and extra < 48:
+            while tgt > 0 and extra < 64:
                 el = max(1e-6, _time.perf_counter() - self._ups_t0)
                 if (self.rl_updates - self._ups_base) >= tgt * el:
                     break
