@@ -65,7 +65,7 @@ class BatchedEngine:
                  rl_serve: str = "device", rl_deterministic: bool = False,
                  rl_exact_p99: bool = False, fp32_coeff_eval: bool = False,
                  rl_tr_limit: Optional[int] = None, rl_reserve_cus: int = 16,
-                 rl_target_updates_per_s: float = 200.0,
+                 rl_target_updates_per_s: float = 180.0,
                  tr_cap: int = 262144, arrival_trace=None,
                  subwave: int = 64, **_unused):
         if algo not in ALGOS:
@@ -267,8 +267,9 @@ class BatchedEngine:
         # cycle trains serially until the measured SAC update rate meets
         # this (<=0 disables the controller: throughput mode, training is
         # purely opportunistic).  The standalone hipGraph rate is ~227/s;
-        # 200 keeps ~1.2M events/s alongside (measured tradeoff curve in
-        # profiles/README.md — 150 -> 2.1M ev/s, 0 -> 5.4M ev/s).
+        # 180 keeps >1M events/s alongside (measured curve in
+        # profiles/README.md: 199 -> 0.74M, 188 -> 0.91M, 149 -> 2.1M,
+        # throughput mode -> 5.4M ev/s).
         self._rl_target_ups = float(rl_target_updates_per_s)
         if self.is_rl:
             t["req_flag"] = torch.zeros(R, **i32)

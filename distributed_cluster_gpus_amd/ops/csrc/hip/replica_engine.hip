@@ -2050,6 +2050,143 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
   }
 }
 
+// ---------------- MFMA batched actor forward ----------------
+// Matrix-core path for BATCHED policy evaluation (host-side serving /
+// offline eval / act-batch): Y = relu(X W^T + b) chained through the whole
+// actor on fp32 MFMA tiles (v_mfma_f32_16x16x4_f32 — exact f32, so results
+// match the fmaf-chain serving path modulo accumulation order).  One
+// workgroup of 16 waves owns 16 batch rows; each wave produces one 16-wide
+// column tile per layer; activations ping-pong through padded LDS.
+// Fragment maps (cdna4_isa.md §10): A[i=l&15][k=l>>4], B[k=l>>4][j=l&15],
+// C/D col=l&15, row=(l>>4)*4+reg.
+constexpr int MF_STRIDE = 257;  // 16-row LDS activation stride (bank-spread)
+using mf_acc = __attribute__((ext_vector_type(4))) float;
+
+__device__ void mfma_layer(const float* Wt, const float* b, int in, int out,
+                           const float* X, float* Y, int wave, int lane,
+                           bool relu, int rows) {
+  int jt = wave * 16;
+  if (jt < out) {
+    mf_acc acc = {0.f, 0.f, 0.f, 0.f};
+    int i = lane & 15, kw = lane >> 4, j = jt + (lane & 15);
+    for (int k0 = 0; k0 < in; k0 += 4) {
+      int k = k0 + kw;
+      float a = (k < in) ? X[i * MF_STRIDE + k] : 0.0f;
+      float w = (k < in && j < out) ? Wt[k * out + j] : 0.0f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, w, acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = (lane >> 4) * 4 + r;
+      if (j < out && row < rows) {
+        float y = acc[r] + b[j];
+        Y[row * MF_STRIDE + j] = relu ? fmaxf(y, 0.0f) : y;
+      }
+    }
+  }
+  __syncthreads();
+}
+
+__global__ void __launch_bounds__(1024)
+rl_forward_mfma_kernel(const float* pw, const float* obs, int B, int obs_dim,
+                       int hid, int n_dc, int n_g, float* out_dc,
+                       float* out_g) {
+  __shared__ float bufA[16 * MF_STRIDE];
+  __shared__ float bufB[16 * MF_STRIDE];
+  int wave = threadIdx.x / 64;
+  int lane = threadIdx.x & 63;
+  int row0 = blockIdx.x * 16;
+  int rows = min(16, B - row0);
+  if (rows <= 0) return;
+  // stage the obs tile (zero-padded)
+  for (int idx = threadIdx.x; idx < 16 * MF_STRIDE; idx += blockDim.x)
+    bufA[idx] = 0.0f;
+  __syncthreads();
+  for (int idx = threadIdx.x; idx < rows * obs_dim; idx += blockDim.x) {
+    int r = idx / obs_dim, k = idx % obs_dim;
+    bufA[r * MF_STRIDE + k] = obs[(int64_t)(row0 + r) * obs_dim + k];
+  }
+  __syncthreads();
+  const int H = hid, D = obs_dim;
+  const float *w1 = pw, *b1 = w1 + D * H;
+  const float *w2 = b1 + H, *b2 = w2 + H * H;
+  const float *w3 = b2 + H, *b3 = w3 + H * H;
+  const float *hdc1 = b3 + H, *hdc1b = hdc1 + H * H;
+  const float *hdc2 = hdc1b + H, *hdc2b = hdc2 + H * n_dc;
+  const float *hg1 = hdc2b + n_dc, *hg1b = hg1 + H * H;
+  const float *hg2 = hg1b + H, *hg2b = hg2 + H * n_g;
+  mfma_layer(w1, b1, D, H, bufA, bufB, wave, lane, true, rows);
+  mfma_layer(w2, b2, H, H, bufB, bufA, wave, lane, true, rows);
+  mfma_layer(w3, b3, H, H, bufA, bufB, wave, lane, true, rows);  // h3 in B
+  mfma_layer(hdc1, hdc1b, H, H, bufB, bufA, wave, lane, true, rows);
+  // dc logits land in the first n_dc columns of a fresh LDS row set; reuse
+  // bufA's upper half is unsafe (still holding hdc1 output) — write the
+  // small head outputs straight to global from the accumulator instead:
+  {
+    int jt = wave * 16;
+    if (jt < n_dc) {
+      mf_acc acc = {0.f, 0.f, 0.f, 0.f};
+      int i = lane & 15, kw = lane >> 4, j = jt + (lane & 15);
+      for (int k0 = 0; k0 < H; k0 += 4) {
+        int k = k0 + kw;
+        float a = bufA[i * MF_STRIDE + k];
+        float w = (j < n_dc) ? hdc2[k * n_dc + j] : 0.0f;
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, w, acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = (lane >> 4) * 4 + r;
+        if (j < n_dc && row < rows)
+          out_dc[(int64_t)(row0 + row) * n_dc + j] = acc[r] + hdc2b[j];
+      }
+    }
+    __syncthreads();
+  }
+  mfma_layer(hg1, hg1b, H, H, bufB, bufA, wave, lane, true, rows);
+  {
+    int jt = wave * 16;
+    if (jt < n_g) {
+      mf_acc acc = {0.f, 0.f, 0.f, 0.f};
+      int i = lane & 15, kw = lane >> 4, j = jt + (lane & 15);
+      for (int k0 = 0; k0 < H; k0 += 4) {
+        int k = k0 + kw;
+        float a = bufA[i * MF_STRIDE + k];
+        float w = (j < n_g) ? hg2[k * n_g + j] : 0.0f;
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, w, acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = (lane >> 4) * 4 + r;
+        if (j < n_g && row < rows)
+          out_g[(int64_t)(row0 + row) * n_g + j] = acc[r] + hg2b[j];
+      }
+    }
+  }
+}
+
+std::vector<torch::Tensor> rl_forward_mfma(torch::Tensor pw,
+                                           torch::Tensor obs,
+                                           int64_t hid, int64_t n_dc,
+                                           int64_t n_g) {
+  TORCH_CHECK(pw.is_cuda() && obs.is_cuda() && pw.dtype() == torch::kFloat32
+              && obs.dtype() == torch::kFloat32 && obs.is_contiguous());
+  TORCH_CHECK(hid % 16 == 0 && hid <= 256 && n_dc <= 16 && n_g <= 16);
+  int B = obs.size(0), D = obs.size(1);
+  auto out_dc = torch::empty({B, n_dc}, obs.options());
+  auto out_g = torch::empty({B, n_g}, obs.options());
+  int blocks = (B + 15) / 16;
+  hipStream_t stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(rl_forward_mfma_kernel, dim3(blocks), dim3(1024), 0,
+                     stream, pw.data_ptr<float>(), obs.data_ptr<float>(), B,
+                     D, (int)hid, (int)n_dc, (int)n_g,
+                     out_dc.data_ptr<float>(), out_g.data_ptr<float>());
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess)
+    throw std::runtime_error(std::string("rl_forward_mfma_kernel: ") +
+                             hipGetErrorString(e));
+  return {out_dc, out_g};
+}
+
 // standalone batched actor forward (test/verification path): one subwave
 // slot per observation row, same device math as the in-engine serving
 __global__ void __launch_bounds__(THREADS_PER_BLOCK)
@@ -2389,6 +2526,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("masked_active", &dcg::BatchedSimHip::masked_active)
       .def("advance_done", &dcg::BatchedSimHip::advance_done)
       .def("advance_sync", &dcg::BatchedSimHip::advance_sync);
+  m.def("rl_forward_mfma", &dcg::rl_forward_mfma,
+        "MFMA (matrix-core) batched actor forward "
+        "(pw, obs[B,D], hid, n_dc, n_g) -> (logits_dc, logits_g)",
+        py::arg("pw"), py::arg("obs"), py::arg("hid"), py::arg("n_dc"),
+        py::arg("n_g"));
   m.def("rl_forward_debug", &dcg::rl_forward_debug,
         "batched actor forward via the in-kernel serving math "
         "(pw, obs[B,D], hid, n_dc, n_g) -> (logits_dc, logits_g)",
