@@ -370,15 +370,21 @@ class BatchedEngine:
             # (layout documented in replica_engine.hip EngineDesc.pw)
             n_g = int(scenario.policy.max_gpus_per_job)
             H = 256
-            if self._serve_device and not self._actor_is_servable(H):
+            servable = self._actor_is_servable(H)
+            if self._serve_device and not servable:
                 import warnings
                 warnings.warn("injected agent has non-standard actor dims; "
                               "falling back to host policy serving")
                 self._serve_device = False
+            # the flat buffer also feeds the MFMA batched forward used by
+            # the host-serving path (matrix-core policy evaluation)
+            self._mfma_serve = servable and not self._serve_device \
+                and not self._rl_det
             wsize = (obs_dim * H + H) + 2 * (H * H + H) + (H * H + H) + \
                 (H * n_dc + n_dc) + (H * H + H) + (H * n_g + n_g)
-            t["policy_weights"] = torch.zeros(wsize if self._serve_device
-                                              else 1, **f32)
+            t["policy_weights"] = torch.zeros(
+                wsize if (self._serve_device or self._mfma_serve) else 1,
+                **f32)
             self._rl_hid = H
             # default yield point: long enough that the overlapped loop can
             # sustain back-to-back train-graph replays under the advance
@@ -428,7 +434,7 @@ class BatchedEngine:
         }
         self._sim = self._mod.BatchedSimHip(t, cfg)
         self.meter = ThroughputMeter()
-        if self._serve_device:
+        if self._serve_device or getattr(self, "_mfma_serve", False):
             self._build_weight_refs()
             self._refresh_policy_weights()
 
@@ -748,6 +754,18 @@ class BatchedEngine:
                                                 device=self.device)
             t["resp_g"][idx] = torch.as_tensor(gs, dtype=torch.int32,
                                                device=self.device)
+        elif getattr(self, "_mfma_serve", False):
+            # matrix-core batched forward (rl_forward_mfma) + the same
+            # masked Gumbel-max sampling the torch path uses
+            from ..rl.masking import sample_categorical
+            with torch.no_grad():
+                ldc, lg = self._mod.rl_forward_mfma(
+                    t["policy_weights"], obs.contiguous(),
+                    self._rl_hid, n_dc, n_g)
+                a_dc, _ = sample_categorical(ldc, m_dc)
+                a_g, _ = sample_categorical(lg, m_g)
+            t["resp_dc"][idx] = a_dc.to(torch.int32)
+            t["resp_g"][idx] = a_g.to(torch.int32)
         else:
             with torch.no_grad():
                 a = self.rl.select_action_batch(obs, m_dc, m_g)
@@ -810,7 +828,8 @@ class BatchedEngine:
                 # sync-free eager SAC step (no stats, tensorized PID)
                 self.rl.train_step(self.replay.sample(self._rl_batch),
                                    compute_stats=False)
-        if refresh and self._serve_device:
+        if refresh and (self._serve_device or
+                        getattr(self, "_mfma_serve", False)):
             self._refresh_policy_weights()
 
     # ---- in-kernel serving support ----

@@ -325,6 +325,7 @@ def test_chsac_batched_respects_masks():
                         duration=60.0, log_interval=5.0, out_dir=None,
                         seed=3, enable_logs=False, rl_warmup=10**9,
                         events_per_launch=5000, rl_serve="host")
+    eng._mfma_serve = False  # route serving through the interceptable torch path
     checked = {"n": 0}
     inner = eng.rl.select_action_batch
 
@@ -870,3 +871,64 @@ def test_chsac_pinned_policy_parity(tmp_path, inf_rate, trn_rate, elastic,
     assert torch.equal(orep.mask_g[:n].cpu(), grep_.mask_g[:n].cpu()), \
         "g-mask diverged (p99-vs-SLA cap path)"
     assert torch.allclose(orep.s[:n].cpu(), grep_.s[:n].cpu(), atol=1e-4)
+
+
+@needs_gpu
+def test_mfma_actor_forward_matches_torch():
+    """The MFMA (matrix-core) batched actor forward must match the torch
+    actor's logits within fp32 accumulation tolerance — fp32 MFMA is exact
+    f32, so only summation order differs."""
+    from distributed_cluster_gpus_amd.ops import load_sim_hip
+    from distributed_cluster_gpus_amd.rl.agent import (CHSACAgent,
+                                                       CHSACAgentConfig)
+    mod = load_sim_hip()
+    torch.manual_seed(17)
+    obs_dim, n_dc, n_g, H = 49, 8, 8, 256
+    agent = CHSACAgent(CHSACAgentConfig(
+        obs_dim=obs_dim, n_dc=n_dc, n_g_choices=n_g,
+        constraints={"latency_p99": 500.0}, device="cuda"))
+    segs = []
+    enc = agent.encoder.net
+    for lin in (enc[0], enc[2], enc[4], agent.actor.head_dc[0],
+                agent.actor.head_dc[2], agent.actor.head_g[0],
+                agent.actor.head_g[2]):
+        segs.append(lin.weight.detach().t().contiguous().reshape(-1))
+        segs.append(lin.bias.detach().reshape(-1))
+    pw = torch.cat(segs).float().cuda()
+    for B in (1, 16, 250, 1024):   # incl. non-multiple-of-16 batch
+        obs = torch.randn(B, obs_dim, device="cuda") * 50.0
+        out_dc, out_g = mod.rl_forward_mfma(pw, obs, H, n_dc, n_g)
+        with torch.no_grad():
+            ref_dc, ref_g = agent.actor(agent.encoder(obs))
+        for got, ref in ((out_dc, ref_dc), (out_g, ref_g)):
+            diff = (got - ref).abs().max().item()
+            scale = ref.abs().max().item() + 1.0
+            assert diff <= 2e-4 * scale, f"B={B}: mfma diff {diff}"
+    # and it agrees with the in-kernel serving forward (same buffer)
+    obs = torch.randn(64, obs_dim, device="cuda")
+    m_dc, m_g2 = mod.rl_forward_mfma(pw, obs, H, n_dc, n_g)
+    d_dc, d_g = mod.rl_forward_debug(pw, obs, H, n_dc, n_g)
+    assert (m_dc - d_dc).abs().max().item() < 1e-3
+    assert (m_g2 - d_g).abs().max().item() < 1e-3
+
+
+@needs_gpu
+def test_chsac_host_serving_uses_mfma_forward():
+    """Non-deterministic host serving routes the policy forward through the
+    matrix-core kernel (production MFMA path), and the run behaves."""
+    from distributed_cluster_gpus_amd.configs.paper import paper_scenario
+    from distributed_cluster_gpus_amd.engine.batched import BatchedEngine
+    from distributed_cluster_gpus_amd.models.arrivals import ArrivalProcess
+    sc = paper_scenario()
+    inf = ArrivalProcess(mode="poisson", rate=1.5)
+    trn = ArrivalProcess(mode="poisson", rate=0.2)
+    eng = BatchedEngine(sc, inf, trn, algo="chsac_af", replicas=24,
+                        duration=90.0, log_interval=5.0, out_dir=None,
+                        seed=13, enable_logs=False, rl_serve="host",
+                        rl_warmup=128, rl_batch=64, rl_train_interval=64,
+                        events_per_launch=5000)
+    assert eng._mfma_serve, "MFMA serving not enabled on the host path"
+    st = eng.run()
+    assert st["jobs_completed"] > 0
+    assert eng.rl_updates > 0
+    assert int(eng.t["err"].max().item()) == 0
