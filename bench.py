@@ -159,6 +159,10 @@ def main():
     duration = max(1200.0, total_steps * args.events_per_step / 100.0)
     qcap = int(max(24576, 8 * duration))
 
+    if args.algo == "chsac_af":
+        raise SystemExit("chsac_af is measured by the RL-in-the-loop metric "
+                         "(rl_loop_events_per_sec, reported with --with-rl 1);"
+                         " the raw step loop would bypass training/ingest")
     sc = paper_scenario()
     inf, trn = build_arrivals()  # sinusoid 6/s amp .6 period 300; poisson 0.3/s
     eng = BatchedEngine(sc, inf, trn, algo=args.algo,

@@ -92,6 +92,9 @@ def parse_args(argv=None):
                    help="Exact 2048-sample sliding-window p99 on the batched "
                         "engine (parity mode) instead of the histogram "
                         "approximation.")
+    p.add_argument("--rl-target-ups", type=float, default=180.0,
+                   help="chsac_af batched engine: SAC update-rate target for "
+                        "the overlapped loop (<=0 = pure-throughput mode).")
     p.add_argument("--fp32-coeff-eval", action="store_true", default=False,
                    help="fp32 decision-score evaluation in the sim kernels "
                         "(batched engine; times/energies stay f64).")
@@ -165,6 +168,7 @@ def main(argv=None):
         eng = BatchedEngine(sc, arrival_inf, arrival_trn,
                             replicas=args.replicas, rl_serve=args.rl_serve,
                             rl_exact_p99=args.rl_exact_p99,
+                            rl_target_updates_per_s=args.rl_target_ups,
                             fp32_coeff_eval=args.fp32_coeff_eval, **common)
     else:
         from distributed_cluster_gpus_amd.engine.oracle import OracleEngine
