@@ -157,7 +157,10 @@ def main():
     # duration generous enough that no replica reaches end_time mid-bench
     # (~150-200 events per simulated second per replica on this workload)
     duration = max(1200.0, total_steps * args.events_per_step / 100.0)
-    qcap = int(max(24576, 8 * duration))
+    # queue capacity: generous default; at huge replica counts the per-entry
+    # f64 pair (size + enqueue time) dominates HBM, so scale it down — paper
+    # workload queues stay in the hundreds (288 GB sizing note)
+    qcap = int(max(24576 if total_replicas <= 16384 else 8192, 8 * duration))
 
     if args.algo == "chsac_af":
         raise SystemExit("chsac_af is measured by the RL-in-the-loop metric "
