@@ -15,7 +15,8 @@ from .render import emit
 
 COMPARISON_FIGURES = [
     "total_power", "cumulative_energy", "utilization", "queues",
-    "latency_hist", "latency_box", "latency_dist",
+    "latency_hist", "latency_box", "latency_violin", "latency_boxen",
+    "latency_dist",
     "energy_vs_latency", "total_energy_bar", "throughput",
     "energy_per_unit_bar", "avg_latency_bar", "completed_jobs_bar",
 ]
@@ -65,7 +66,21 @@ def comparison_report(run_dirs: Dict[str, str], out_dir: str) -> List[str]:
     arts.append(emit(inf[["latency_s", "algo"]], out_dir, "latency_box",
                      kind="box", y="latency_s", hue="algo",
                      title="Inference latency distribution (box)"))
-    # distribution view (quantile curves — the violin/boxen analog as data)
+    # violin + letter-value (boxen) views, matching the reference's seaborn
+    # figures (plot_sim_result.py latency views); raw values are sampled to
+    # bound the CSV artifact size
+    lat = inf[["latency_s", "algo"]]
+    if len(lat) > 50000:
+        lat = lat.groupby("algo", group_keys=False).apply(
+            lambda g: g.sample(min(len(g), 50000 // max(1, lat["algo"].nunique())),
+                               random_state=0))
+    arts.append(emit(lat, out_dir, "latency_violin", kind="violin",
+                     y="latency_s", hue="algo",
+                     title="Inference latency distribution (violin)"))
+    arts.append(emit(lat, out_dir, "latency_boxen", kind="boxen",
+                     y="latency_s", hue="algo",
+                     title="Inference latency distribution (letter-value)"))
+    # distribution view (quantile curves as data)
     rows = []
     for label, sub in inf.groupby("algo"):
         qs = np.linspace(0.01, 0.99, 99)

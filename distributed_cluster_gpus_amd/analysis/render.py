@@ -7,6 +7,7 @@ matplotlib (this ROCm image ships pandas but not matplotlib)."""
 import os
 from typing import Optional
 
+import numpy as np
 import pandas as pd
 
 
@@ -30,7 +31,7 @@ def emit(df: pd.DataFrame, out_dir: str, name: str,
          y=None, title: str = "", ylabel: str = "",
          hue: Optional[str] = None, logy: bool = False) -> str:
     """Write <name>.csv always; render <name>.png when matplotlib exists.
-    kind: line | bar | hist | scatter | heatmap | box."""
+    kind: line | bar | hist | scatter | heatmap | box | violin | boxen."""
     os.makedirs(out_dir, exist_ok=True)
     csv_path = os.path.join(out_dir, f"{name}.csv")
     df.to_csv(csv_path, index=False)
@@ -76,6 +77,39 @@ def emit(df: pd.DataFrame, out_dir: str, name: str,
                 ax.boxplot([g[1] for g in groups],
                            labels=[g[0] for g in groups], showfliers=False)
                 ax.tick_params(axis="x", rotation=30)
+        elif kind == "violin":
+            groups = [(str(k), sub[y].dropna().values)
+                      for k, sub in df.groupby(hue) if len(sub)]
+            if groups:
+                ax.violinplot([g[1] for g in groups], showmedians=True,
+                              showextrema=False)
+                ax.set_xticks(range(1, len(groups) + 1))
+                ax.set_xticklabels([g[0] for g in groups], rotation=30)
+        elif kind == "boxen":
+            # letter-value plot: nested boxes at halving tail depths
+            # (reference renders this via seaborn.boxenplot; same construct)
+            from matplotlib.patches import Rectangle
+            groups = [(str(k), np.sort(sub[y].dropna().values))
+                      for k, sub in df.groupby(hue) if len(sub)]
+            for i, (label, vals) in enumerate(groups, start=1):
+                if len(vals) < 2:
+                    continue
+                depth, width = 0.25, 0.8
+                while depth * len(vals) >= 1 and width > 0.05:
+                    lo = float(np.quantile(vals, depth))
+                    hi = float(np.quantile(vals, 1 - depth))
+                    ax.add_patch(Rectangle((i - width / 2, lo), width,
+                                           max(hi - lo, 1e-12),
+                                           facecolor="C%d" % ((i - 1) % 10),
+                                           alpha=0.35, edgecolor="none"))
+                    depth /= 2
+                    width *= 0.7
+                med = float(np.median(vals))
+                ax.plot([i - 0.4, i + 0.4], [med, med], color="k", lw=1.2)
+            if groups:
+                ax.set_xlim(0.4, len(groups) + 0.6)
+                ax.set_xticks(range(1, len(groups) + 1))
+                ax.set_xticklabels([g[0] for g in groups], rotation=30)
         elif kind == "heatmap":
             mat = df.set_index(df.columns[0])
             im = ax.imshow(mat.values, aspect="auto", cmap="viridis")
