@@ -155,31 +155,20 @@ class BatchedEngine:
         t["dc_min_finish"] = torch.full((R, n_dc), INF, **f64)
         t["dc_min_slot"] = torch.full((R, n_dc), -1, **i32)
         t["s_finish"] = torch.full((R, total_slots), INF, **f64)
-        t["s_start"] = torch.zeros((R, total_slots), **f64)
-        t["s_lastupd"] = torch.zeros((R, total_slots), **f64)
-        t["s_seq"] = torch.zeros((R, total_slots), **i32)
+        # packed cold slot payload (SRec, 64 B — start/lastupd/size/fused/
+        # done f64, netlat f32, jid/seq i32, ing/pcount/RL-trace bytes);
+        # one cache line per job start/finish instead of 10+ SoA touches
+        t["s_rec"] = torch.zeros((R, total_slots, 8), **f64)
         t["seq_ctr"] = torch.zeros(R, **i32)
-        t["s_size"] = torch.zeros((R, total_slots), **f64)
-        t["s_fused"] = torch.zeros((R, total_slots), **f64)
-        t["s_netlat"] = torch.zeros((R, total_slots), **f32)
-        t["s_jid"] = torch.zeros((R, total_slots), **i32)
         t["s_gpus"] = torch.zeros((R, total_slots), **i16)
         t["s_jtype"] = torch.zeros((R, total_slots), **i8)
-        t["s_ing"] = torch.zeros((R, total_slots), **i8)
-        t["s_done"] = torch.zeros((R, total_slots), **f64)
-        t["s_pcount"] = torch.zeros((R, total_slots),
-                                    dtype=torch.uint8, device=dev)
         t["x_time"] = torch.full((R, tcap), INF, **f64)
-        t["x_size"] = torch.zeros((R, tcap), **f64)
-        t["x_netlat"] = torch.zeros((R, tcap), **f32)
-        t["x_jid"] = torch.zeros((R, tcap), **i32)
-        t["x_dc"] = torch.zeros((R, tcap), **i8)
-        t["x_jtype"] = torch.zeros((R, tcap), **i8)
-        t["x_ing"] = torch.zeros((R, tcap), **i8)
+        # packed transfer payload (XRec, 32 B)
+        t["x_rec"] = torch.zeros((R, tcap, 4), **f64)
         t["q_head"] = torch.zeros((R, n_dc, 2), **i32)
         t["q_len"] = torch.zeros((R, n_dc, 2), **i32)
-        t["q_size"] = torch.zeros((R, n_dc, 2, qcap), **f64)
-        t["q_enq"] = torch.zeros((R, n_dc, 2, qcap), **f64)
+        # per-entry (size, enqueue-time) f64 pair — one line per push/pop
+        t["q_pay"] = torch.zeros((R, n_dc, 2, qcap, 2), **f64)
         # cap_greedy pass-snapshot scratch (frozen task frequencies)
         t["snap_f"] = torch.zeros(
             (R, total_slots) if algo == "cap_greedy" else (1, 1), **f64)
@@ -288,20 +277,9 @@ class BatchedEngine:
             t["pend_from_inf"] = torch.zeros(R, **i32)
             t["pend_enq"] = torch.zeros(R, **f64)
             u8 = dict(dtype=torch.uint8, device=dev)
+            # per-job RL obs traces; action/mask bytes live in s_rec/x_rec
             t["slot_s0"] = torch.zeros((R, total_slots, obs_dim), **f32)
-            t["slot_adc"] = torch.zeros((R, total_slots), **u8)
-            t["slot_ag"] = torch.zeros((R, total_slots), **u8)
-            t["slot_mdc"] = torch.zeros((R, total_slots), **u8)
-            t["slot_mg"] = torch.zeros((R, total_slots), **u8)
-            t["slot_has_rl"] = torch.zeros((R, total_slots), **u8)
-            t["slot_nrew"] = torch.zeros((R, total_slots), **u8)
             t["x_s0"] = torch.zeros((R, tcap, obs_dim), **f32)
-            t["x_nsel"] = torch.zeros((R, tcap), **i16)
-            t["x_adc"] = torch.zeros((R, tcap), **u8)
-            t["x_ag"] = torch.zeros((R, tcap), **u8)
-            t["x_mdc"] = torch.zeros((R, tcap), **u8)
-            t["x_mg"] = torch.zeros((R, tcap), **u8)
-            t["x_has_rl"] = torch.zeros((R, tcap), **u8)
             # elastic-scaling preempted-job pool: a DC can run up to
             # total_gpus concurrent 1-GPU training jobs, all preemptible
             pp_cap = int(scenario.total_gpus.max())
@@ -965,7 +943,7 @@ class BatchedEngine:
         assert torch.equal(run_ref, t["n_running"].long()),             "n_running != active slot count"
         # busy never exceeds capacity; queue lengths within bounds
         assert bool((t["busy"] <= t["total_gpus"].unsqueeze(0)).all()),             "busy > total_gpus"
-        assert bool((t["q_len"] >= 0).all()) and             bool((t["q_len"] < t["q_size"].shape[-1]).all() or True)
+        assert bool((t["q_len"] >= 0).all()) and             bool((t["q_len"] < t["q_pay"].shape[-2]).all() or True)
         # empty slots hold +inf finish; active slots hold finite times
         fin = t["s_finish"]
         assert bool((fin[~active] >= INF).all()), "empty slot with finite finish"

@@ -40,6 +40,32 @@ namespace dcg {
 constexpr int MAX_DC = 8;
 constexpr int MAX_FREQ = 8;
 
+// Packed per-object payload records: everything a job start/finish (or a
+// WAN-transfer push/admission) touches lands in ONE cache line instead of
+// 6-12 scattered SoA arrays — the engine is memory-LATENCY-chain bound
+// (profiles/README.md PMC analysis), so dependent line count per event is
+// the cost that matters.  Scan-heavy fields (finish times, gpus, jtype)
+// stay SoA/LDS.
+struct XRec {            // 32 B: in-flight WAN transfer payload
+  double size;
+  float netlat;
+  int jid;
+  short nsel;            // RL-chosen n (chsac; unclamped g+1)
+  unsigned char dc, jtype, ing, adc, ag, mdc;
+  unsigned char mg, has_rl, pad0, pad1;
+};
+static_assert(sizeof(XRec) == 32, "XRec must be one 32B record");
+
+struct SRec {            // 64 B: cold job-slot payload
+  double start, lastupd, size, fused, done;
+  float netlat;
+  int jid;
+  int seq;
+  unsigned char ing, pcount, nrew, adc, ag, mdc, mg, has_rl;
+  unsigned char pad[4];
+};
+static_assert(sizeof(SRec) == 64, "SRec must be one 64B record");
+
 enum Algo { A_DEFAULT = 0, A_CAP_UNIFORM, A_CAP_GREEDY, A_JOINT_NF, A_BANDIT,
             A_CARBON_COST, A_ECO_ROUTE, A_DEBUG, A_CHSAC };
 
@@ -108,38 +134,21 @@ struct EngineDesc {
   int* n_running;
   double* dc_min_finish;          // cached min finish time (INF if none)
   int* dc_min_slot;
-  // job slots [r][total_slots]
+  // job slots [r][total_slots]: finish times + scan fields SoA, the cold
+  // payload as one SRec per slot (f64 values bitwise-exact as before)
   double* s_finish;               // INF = empty
-  double* s_start;
-  double* s_size;                 // f64: service times must be bitwise-equal
-  double* s_fused;                //   to the scalar engines (exact parity)
-  float* s_netlat;
-  int* s_jid;
-  short* s_gpus;                  // 0 = empty
-  char* s_jtype;
-  char* s_ing;
-  double* s_done;                 // units completed before this (re)start
-  double* s_lastupd;              // last progress-update time (reference
-                                  //   Job.last_update; advanced by the cap
-                                  //   controller's reschedule)
-  int* s_seq;                     // monotone start sequence — running_jobs
-                                  //   dict insertion order (atom tie-break)
-  unsigned char* s_pcount;        // preemptions experienced
-  // in-flight transfers [r][tcap]
+  SRec* s_rec;                    // packed payload
+  short* s_gpus;                  // 0 = empty (dense busy/log scans)
+  char* s_jtype;                  // dense scans (log rows, elastic, cap)
+  // in-flight transfers [r][tcap]: time SoA (LDS-mirrored), payload packed
   double* x_time;                 // INF = empty
-  double* x_size;
-  float* x_netlat;
-  int* x_jid;
-  char* x_dc;
-  char* x_jtype;
-  char* x_ing;
-  // queues [r][dc][2] ring of capacity qcap; aux fields (netlat/jid/ing)
-  // exist only for the logging replica ([dc][2][qcap], no replica dim)
+  XRec* x_rec;
+  // queues [r][dc][2] ring of capacity qcap; per-entry payload = (size,
+  // enqueue-time) f64 pair; aux fields (netlat/jid/ing) exist only for the
+  // logging replica ([dc][2][qcap], no replica dim)
   int* q_head;
   int* q_len;
-  double* q_size;                 // [r][dc][2][qcap] (f64 for exact parity)
-  double* q_enq;                  // [r][dc][2][qcap] enqueue time (queueing-
-                                  //   delay metric; FIFO-paired with pops)
+  double* q_pay;                  // [r][dc][2][qcap][2] (f64 exact parity)
   float* q_netlat;                // [dc][2][qcap] (log replica only)
   int* q_jid;                     // [dc][2][qcap] (log replica only)
   char* q_ing;                    // [dc][2][qcap] (log replica only)
@@ -212,23 +221,11 @@ struct EngineDesc {
   int* pend_dc;                   // [r] (drain: source DC)
   int* pend_from_inf;             // [r] (drain: which queue the job came from)
   double* pend_enq;               // [r] (drain: the popped job's enqueue time)
-  // per-job RL traces (state0 / action / masks at selection time)
+  // per-job RL obs traces (the action/mask bytes live in SRec/XRec;
+  // arrival path keeps nrew = g_idx+1 unclamped, drain path clamped —
+  // reference :571 vs :889)
   float* slot_s0;                 // [r][total_slots][obs_dim]
-  unsigned char* slot_adc;        // [r][total_slots]
-  unsigned char* slot_ag;         // [r][total_slots]
-  unsigned char* slot_mdc;        // [r][total_slots]
-  unsigned char* slot_mg;         // [r][total_slots]
-  unsigned char* slot_has_rl;     // [r][total_slots]
-  unsigned char* slot_nrew;       // [r][total_slots] n used in the reward
-                                  // (arrival path: g_idx+1 unclamped, drain
-                                  // path: clamped — reference :571 vs :889)
   float* x_s0;                    // [r][tcap][obs_dim]
-  short* x_nsel;                  // [r][tcap] RL-chosen n (before clamping)
-  unsigned char* x_adc;           // [r][tcap]
-  unsigned char* x_ag;            // [r][tcap]
-  unsigned char* x_mdc;           // [r][tcap]
-  unsigned char* x_mg;            // [r][tcap]
-  unsigned char* x_has_rl;        // [r][tcap]
   // elastic scaling (chsac only): preempted-training-job pool per replica
   int elastic;                    // 0 off, 1 on
   int pp_cap;                     // pool capacity
@@ -401,19 +398,20 @@ __device__ void start_job(Ctx& c, int d, int jt, double size, float netlat,
   double finish = now + (double)size * T;
   if (c.lane == 0) {
     c.l_fin[cand] = finish;
-    S.s_start[base + cand] = now;
-    S.s_lastupd[base + cand] = now;
-    S.s_size[base + cand] = size;
-    S.s_fused[base + cand] = f;
-    S.s_netlat[base + cand] = netlat;
-    S.s_jid[base + cand] = jid;
+    SRec* sr = &S.s_rec[base + cand];
+    sr->start = now;
+    sr->lastupd = now;
+    sr->size = size;
+    sr->fused = f;
+    sr->netlat = netlat;
+    sr->jid = jid;
+    sr->done = 0.0;
+    sr->pcount = 0;
+    sr->ing = (unsigned char)ing;
+    sr->seq = ++c.hs->seq_ctr;
+    sr->has_rl = 0;
     S.s_gpus[base + cand] = (short)n;
     S.s_jtype[base + cand] = (char)jt;
-    S.s_ing[base + cand] = (char)ing;
-    S.s_done[base + cand] = 0.0;
-    S.s_pcount[base + cand] = 0;
-    int sq = ++c.hs->seq_ctr;
-    S.s_seq[base + cand] = sq;
     c.hs->busy[d] += n;
     c.hs->n_running[d] += 1;
     c.hs->p_active[d] += d_job_power(n, f, c.pc3(d, jt));
@@ -521,8 +519,8 @@ __device__ bool queue_push(Ctx& c, int d, int jt, double size, float netlat,
   int pos = (c.hs->q_head[ql] + len) % S.qcap;
   if (c.lane == 0) {
     int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
-    S.q_size[at] = size;
-    S.q_enq[at] = enq;
+    S.q_pay[at * 2] = size;
+    S.q_pay[at * 2 + 1] = enq;
     if (c.r == S.log_replica) {
       int64_t aux = ((int64_t)ql) * S.qcap + pos;
       S.q_netlat[aux] = netlat;
@@ -547,8 +545,8 @@ __device__ bool queue_push_front(Ctx& c, int d, int jt, double size,
   int pos = (c.hs->q_head[ql] - 1 + S.qcap) % S.qcap;
   if (c.lane == 0) {
     int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
-    S.q_size[at] = size;
-    S.q_enq[at] = enq;
+    S.q_pay[at * 2] = size;
+    S.q_pay[at * 2 + 1] = enq;
     if (c.r == S.log_replica) {
       int64_t aux = ((int64_t)ql) * S.qcap + pos;
       S.q_netlat[aux] = netlat;
@@ -570,8 +568,8 @@ __device__ bool queue_pop(Ctx& c, int d, int jt, double& size, float& netlat,
   if (len <= 0) return false;
   int pos = c.hs->q_head[ql];
   int64_t at = ((int64_t)(c.r * S.n_dc + d) * 2 + jt) * S.qcap + pos;
-  size = S.q_size[at];
-  enq = S.q_enq[at];
+  size = S.q_pay[at * 2];
+  enq = S.q_pay[at * 2 + 1];
   if (c.r == S.log_replica) {
     int64_t aux = ((int64_t)ql) * S.qcap + pos;
     netlat = S.q_netlat[aux];
@@ -744,7 +742,7 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
     for (int k = c.lane; k < S.total_slots; k += SUBW) {
       double sf = 0.0;
       if (c.l_fin[k] < D_INF) {
-        double fu = S.s_fused[base + k];
+        double fu = S.s_rec[base + k].fused;
         if (fu > f_min + 1e-12) { sf = fu; have = 1; }
       }
       S.snap_f[base + k] = sf;
@@ -773,7 +771,7 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
         int d = S.slot_dc[k];
         int jt = S.s_jtype[base + k];
         int n = S.s_gpus[base + k];
-        long long seq = S.s_seq[base + k];
+        long long seq = S.s_rec[base + k].seq;
         double curV = 1.0 / d_unit_time(n, lv[i0], c.lc3(d, jt));
         double curP = d_job_power(n, lv[i0], c.pc3(d, jt));
         for (int kk = i0; kk >= 1; --kk) {
@@ -808,7 +806,8 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
       last_rho = best_rho;
       last_key = best_key;
       // skip unless the atom still goes strictly DOWN from the live f
-      double cur_f = S.s_fused[base + best_slot];
+      SRec* bs = &S.s_rec[base + best_slot];
+      double cur_f = bs->fused;
       double f_to = lv[best_k - 1];
       if (f_to >= cur_f - 1e-12) continue;
       // ---- apply: the reference's exact reschedule arithmetic ----
@@ -817,17 +816,17 @@ __device__ void cap_greedy_control(Ctx& c, double now) {
       int n = S.s_gpus[base + best_slot];
       double T_old = d_unit_time(n, cur_f, c.lc3(d, jt));
       double rate_old = 1.0 / fmax(T_old, 1e-9);
-      double size = S.s_size[base + best_slot];
-      double dt = fmax(0.0, now - S.s_lastupd[base + best_slot]);
-      double done = fmin(size, S.s_done[base + best_slot] + rate_old * dt);
+      double size = bs->size;
+      double dt = fmax(0.0, now - bs->lastupd);
+      double done = fmin(size, bs->done + rate_old * dt);
       double units_left = fmax(0.0, size - done);
       double T_new = d_unit_time(n, f_to, c.lc3(d, jt));
       double rate_new = 1.0 / fmax(T_new, 1e-9);
       double finish_new = now + units_left / fmax(rate_new, 1e-9);
       if (c.lane == 0) {
-        S.s_done[base + best_slot] = done;
-        S.s_lastupd[base + best_slot] = now;
-        S.s_fused[base + best_slot] = f_to;
+        bs->done = done;
+        bs->lastupd = now;
+        bs->fused = f_to;
         c.l_fin[best_slot] = finish_new;
         c.hs->p_active[d] += d_job_power(n, f_to, c.pc3(d, jt)) -
                              d_job_power(n, cur_f, c.pc3(d, jt));
@@ -1133,25 +1132,25 @@ __device__ void rl_start_job(Ctx& c, int d, int jt, double size, float netlat,
     S.slot_s0[(base + cand) * S.obs_dim + k] = s0[k];
   if (c.lane == 0) {
     c.l_fin[cand] = finish;
-    S.s_start[base + cand] = orig_start >= 0.0 ? orig_start : now;
-    S.s_lastupd[base + cand] = now;
-    S.s_size[base + cand] = size;
-    S.s_fused[base + cand] = f;
-    S.s_netlat[base + cand] = netlat;
-    S.s_jid[base + cand] = jid;
+    SRec* sr = &S.s_rec[base + cand];
+    sr->start = orig_start >= 0.0 ? orig_start : now;
+    sr->lastupd = now;
+    sr->size = size;
+    sr->fused = f;
+    sr->netlat = netlat;
+    sr->jid = jid;
+    sr->done = units_done;
+    sr->pcount = (unsigned char)pcount;
+    sr->ing = (unsigned char)ing;
+    sr->seq = ++c.hs->seq_ctr;
+    sr->adc = (unsigned char)a_dc;
+    sr->ag = (unsigned char)a_g;
+    sr->mdc = (unsigned char)mdc;
+    sr->mg = (unsigned char)mg;
+    sr->has_rl = (unsigned char)has_rl;
+    sr->nrew = (unsigned char)max(1, n_rew);
     S.s_gpus[base + cand] = (short)n;
     S.s_jtype[base + cand] = (char)jt;
-    S.s_ing[base + cand] = (char)ing;
-    S.s_done[base + cand] = units_done;
-    S.s_pcount[base + cand] = (unsigned char)pcount;
-    int sq = ++c.hs->seq_ctr;
-    S.s_seq[base + cand] = sq;
-    S.slot_adc[base + cand] = (unsigned char)a_dc;
-    S.slot_ag[base + cand] = (unsigned char)a_g;
-    S.slot_mdc[base + cand] = (unsigned char)mdc;
-    S.slot_mg[base + cand] = (unsigned char)mg;
-    S.slot_has_rl[base + cand] = (unsigned char)has_rl;
-    S.slot_nrew[base + cand] = (unsigned char)max(1, n_rew);
     c.hs->busy[d] += n;
     c.hs->n_running[d] += 1;
     c.hs->p_active[d] += d_job_power(n, f, c.pc3(d, jt));
@@ -1214,7 +1213,7 @@ __device__ int rl_elastic_preempt_all(Ctx& c, int d, double now) {
     int k = -1, best_seq = INT_MAX;
     for (int q = lo; q < hi; ++q) {
       if (S.s_gpus[base + q] == 0 || S.s_jtype[base + q] != 1) continue;
-      int sq = S.s_seq[base + q];
+      int sq = S.s_rec[base + q].seq;
       if (sq < best_seq) { best_seq = sq; k = q; }
     }
     if (k < 0) break;
@@ -1223,26 +1222,27 @@ __device__ int rl_elastic_preempt_all(Ctx& c, int d, double now) {
       break;
     }
     int n = S.s_gpus[base + k];
-    double f = S.s_fused[base + k];
+    SRec* sk = &S.s_rec[base + k];
+    double f = sk->fused;
     double T = d_unit_time(n, f, c.lc3(d, 1));
-    double done = S.s_done[base + k] +
-                  fmax(0.0, now - S.s_lastupd[base + k]) / fmax(T, 1e-300);
-    double size = S.s_size[base + k];
+    double done = sk->done +
+                  fmax(0.0, now - sk->lastupd) / fmax(T, 1e-300);
+    double size = sk->size;
     done = fmin(size, done);
     int64_t pb = (int64_t)c.r * S.pp_cap + count;
     if (c.lane == 0) {
       S.pp_size[pb] = size;
       S.pp_done[pb] = done;
-      S.pp_start[pb] = S.s_start[base + k];
-      S.pp_netlat[pb] = S.s_netlat[base + k];
-      S.pp_jid[pb] = S.s_jid[base + k];
-      S.pp_ing[pb] = (unsigned char)S.s_ing[base + k];
+      S.pp_start[pb] = sk->start;
+      S.pp_netlat[pb] = sk->netlat;
+      S.pp_jid[pb] = sk->jid;
+      S.pp_ing[pb] = sk->ing;
       S.pp_dc[pb] = (unsigned char)d;
-      S.pp_pcount[pb] = (unsigned char)(S.s_pcount[base + k] + 1);
-      S.pp_adc[pb] = S.slot_adc[base + k];
-      S.pp_ag[pb] = S.slot_ag[base + k];
-      S.pp_nrew[pb] = S.slot_nrew[base + k];
-      S.pp_has_rl[pb] = S.slot_has_rl[base + k];
+      S.pp_pcount[pb] = (unsigned char)(sk->pcount + 1);
+      S.pp_adc[pb] = sk->adc;
+      S.pp_ag[pb] = sk->ag;
+      S.pp_nrew[pb] = sk->nrew;
+      S.pp_has_rl[pb] = sk->has_rl;
       // free the slot + caches
       S.s_gpus[base + k] = 0;
       c.l_fin[k] = D_INF;
@@ -1329,18 +1329,19 @@ __device__ void rl_do_arrival(Ctx& c, double now, int jt, int ing,
     S.x_s0[at * S.obs_dim + k] = s0[k];
   if (c.lane == 0) {
     c.l_xt[cand] = now + lnet + xfer;
-    S.x_size[at] = size;
-    S.x_netlat[at] = (float)lnet;
-    S.x_jid[at] = jid;
-    S.x_dc[at] = (char)d_sel;
-    S.x_jtype[at] = (char)jt;
-    S.x_ing[at] = (char)ing;
-    S.x_nsel[at] = (short)n_sel;
-    S.x_adc[at] = (unsigned char)a_dc;
-    S.x_ag[at] = (unsigned char)a_g;
-    S.x_mdc[at] = (unsigned char)mdc;
-    S.x_mg[at] = (unsigned char)mg;
-    S.x_has_rl[at] = 1;
+    XRec* xr = &S.x_rec[at];
+    xr->size = size;
+    xr->netlat = (float)lnet;
+    xr->jid = jid;
+    xr->dc = (unsigned char)d_sel;
+    xr->jtype = (unsigned char)jt;
+    xr->ing = (unsigned char)ing;
+    xr->nsel = (short)n_sel;
+    xr->adc = (unsigned char)a_dc;
+    xr->ag = (unsigned char)a_g;
+    xr->mdc = (unsigned char)mdc;
+    xr->mg = (unsigned char)mg;
+    xr->has_rl = 1;
   }
   store_fence();
 }
@@ -1803,12 +1804,14 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       } else if (lane == 0) {
         int64_t at = (int64_t)c.r * S.tcap + cand;
         c.l_xt[cand] = t_min + lnet + xfer;
-        S.x_size[at] = size;
-        S.x_netlat[at] = (float)lnet;
-        S.x_jid[at] = jid;
-        S.x_dc[at] = (char)d_sel;
-        S.x_jtype[at] = (char)jt;
-        S.x_ing[at] = (char)ing;
+        XRec* xr = &S.x_rec[at];
+        xr->size = size;
+        xr->netlat = (float)lnet;
+        xr->jid = jid;
+        xr->dc = (unsigned char)d_sel;
+        xr->jtype = (unsigned char)jt;
+        xr->ing = (unsigned char)ing;
+        xr->has_rl = 0;
       }
       store_fence();
       // next arrival for this stream (faithful non-accumulating thinning;
@@ -1840,22 +1843,23 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
     } else if (kind == 1) {
       // ===== WAN transfer complete: admission =====
       int64_t at = (int64_t)c.r * S.tcap + idx;
-      int d = S.x_dc[at];
-      int jt = S.x_jtype[at];
-      double size = S.x_size[at];
-      float netlat = S.x_netlat[at];
-      int jid = S.x_jid[at];
-      int ing = S.x_ing[at];
+      XRec xr = S.x_rec[at];
+      int d = xr.dc;
+      int jt = xr.jtype;
+      double size = xr.size;
+      float netlat = xr.netlat;
+      int jid = xr.jid;
+      int ing = xr.ing;
       if (lane == 0) c.l_xt[idx] = D_INF;
       lds_fence();
       if (c.free_gpus(d) > 0) {
-        if (ALGO == A_CHSAC && S.x_has_rl[at]) {
+        if (ALGO == A_CHSAC && xr.has_rl) {
           // RL-chosen n (clamped), energy-optimal f (reference :646-667)
-          int n = max(1, min(min((int)S.x_nsel[at], c.free_gpus(d)), S.max_gpj));
+          int n = max(1, min(min((int)xr.nsel, c.free_gpus(d)), S.max_gpj));
           double f = rl_energy_freq(c, d, jt, n);
           rl_start_job(c, d, jt, size, netlat, jid, ing, n, f, t_min,
-                       &S.x_s0[at * S.obs_dim], S.x_adc[at], S.x_ag[at],
-                       S.x_mdc[at], S.x_mg[at], (int)S.x_nsel[at]);
+                       &S.x_s0[at * S.obs_dim], xr.adc, xr.ag,
+                       xr.mdc, xr.mg, (int)xr.nsel);
         } else {
           int n; double f;
           decide_nf<ALGO>(c, d, jt, size, t_min, n, f);
@@ -1875,13 +1879,14 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
       int64_t at = sbase + slot;
       int jt = S.s_jtype[at];
       int n = S.s_gpus[at];
-      double size = S.s_size[at];
-      double fused = S.s_fused[at];
-      double start = S.s_start[at];
-      float netlat = S.s_netlat[at];
-      int jid = S.s_jid[at];
-      int ingr = S.s_ing[at];
-      int pcount = S.s_pcount[at];
+      SRec srv = S.s_rec[at];
+      double size = srv.size;
+      double fused = srv.fused;
+      double start = srv.start;
+      float netlat = srv.netlat;
+      int jid = srv.jid;
+      int ingr = srv.ing;
+      int pcount = srv.pcount;
       double T = d_unit_time(n, fused, c.lc3(d, jt));
       if (lane == 0) {
         c.l_fin[slot] = D_INF;
@@ -1924,11 +1929,11 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
         double sojourn = fmax(0.0, t_min - start);
         rl_record_latency(c, jt, sojourn);
         store_fence();
-        if (S.slot_has_rl[at]) {
+        if (srv.has_rl) {
           double E_pred = d_job_power(n, fused, c.pc3(d, jt)) * T;  // J/unit
           double E_unit_kwh = (E_pred * (double)size / 3.6e6) /
                               ((double)size + 1e-9);
-          int n_rew = max(1, (int)S.slot_nrew[at]);
+          int n_rew = max(1, (int)srv.nrew);
           float r = (float)(-E_unit_kwh + 0.05 * (1.0 / n_rew));
           double p99 = rl_p99_ms(c, jt);
           if (p99 < 0.0) p99 = sojourn * 1000.0;  // <5 samples fallback
@@ -1938,9 +1943,9 @@ advance_kernel(EngineDesc S, double t_target, long long max_ev) {
           int mdc, mg;
           rl_build_masks(c, mdc, mg);  // masks at completion (reference :793)
           rl_emit_transition(c, &S.slot_s0[at * S.obs_dim],
-                             S.slot_adc[at], S.slot_ag[at], r,
+                             srv.adc, srv.ag, r,
                              (float)p99, (float)P_now, c_over, mdc, mg, t_min);
-          if (lane == 0) S.slot_has_rl[at] = 0;
+          if (lane == 0) S.s_rec[at].has_rl = 0;
           store_fence();
         }
         // elastic scaling: on a training completion with other training jobs
@@ -2300,21 +2305,13 @@ class BatchedSimHip {
     T_PTR(util_time, double); T_PTR(util_begin, double); T_PTR(acc_unit, double);
     T_PTR(p_active, double); T_PTR(sum_tpt, double); T_PTR(n_running, int);
     T_PTR(dc_min_finish, double); T_PTR(dc_min_slot, int);
-    T_PTR(s_finish, double); T_PTR(s_start, double); T_PTR(s_size, double);
-    T_PTR(s_lastupd, double); T_PTR(s_seq, int); T_PTR(seq_ctr, int);
-    T_PTR(s_fused, double); T_PTR(s_netlat, float); T_PTR(s_jid, int);
+    T_PTR(s_finish, double); T_PTR(seq_ctr, int);
+    S_.s_rec = reinterpret_cast<SRec*>(t_["s_rec"].data_ptr<double>());
     S_.s_gpus = reinterpret_cast<short*>(t_["s_gpus"].data_ptr<int16_t>());
     S_.s_jtype = reinterpret_cast<char*>(t_["s_jtype"].data_ptr<int8_t>());
-    S_.s_ing = reinterpret_cast<char*>(t_["s_ing"].data_ptr<int8_t>());
-    T_PTR(s_done, double);
-    S_.s_pcount = reinterpret_cast<unsigned char*>(t_["s_pcount"].data_ptr<uint8_t>());
-    T_PTR(x_time, double); T_PTR(x_size, double); T_PTR(x_netlat, float);
-    T_PTR(x_jid, int);
-    S_.x_dc = reinterpret_cast<char*>(t_["x_dc"].data_ptr<int8_t>());
-    S_.x_jtype = reinterpret_cast<char*>(t_["x_jtype"].data_ptr<int8_t>());
-    S_.x_ing = reinterpret_cast<char*>(t_["x_ing"].data_ptr<int8_t>());
-    T_PTR(q_head, int); T_PTR(q_len, int); T_PTR(q_size, double);
-    T_PTR(q_enq, double);
+    T_PTR(x_time, double);
+    S_.x_rec = reinterpret_cast<XRec*>(t_["x_rec"].data_ptr<double>());
+    T_PTR(q_head, int); T_PTR(q_len, int); T_PTR(q_pay, double);
     if (S_.algo == A_CAP_GREEDY) T_PTR(snap_f, double);
     T_PTR(q_netlat, float); T_PTR(q_jid, int);
     S_.q_ing = reinterpret_cast<char*>(t_["q_ing"].data_ptr<int8_t>());
@@ -2363,12 +2360,6 @@ class BatchedSimHip {
       T_PTR(pend_jid, int); T_PTR(pend_ing, int); T_PTR(pend_jt, int);
       T_PTR(pend_dc, int); T_PTR(pend_from_inf, int); T_PTR(pend_enq, double);
       T_PTR(slot_s0, float);
-      S_.slot_adc = reinterpret_cast<unsigned char*>(t_["slot_adc"].data_ptr<uint8_t>());
-      S_.slot_ag = reinterpret_cast<unsigned char*>(t_["slot_ag"].data_ptr<uint8_t>());
-      S_.slot_mdc = reinterpret_cast<unsigned char*>(t_["slot_mdc"].data_ptr<uint8_t>());
-      S_.slot_mg = reinterpret_cast<unsigned char*>(t_["slot_mg"].data_ptr<uint8_t>());
-      S_.slot_has_rl = reinterpret_cast<unsigned char*>(t_["slot_has_rl"].data_ptr<uint8_t>());
-      S_.slot_nrew = reinterpret_cast<unsigned char*>(t_["slot_nrew"].data_ptr<uint8_t>());
       T_PTR(pp_count, int); T_PTR(pp_cursor, int);
       T_PTR(pp_size, double); T_PTR(pp_done, double); T_PTR(pp_start, double);
       T_PTR(pp_netlat, float);
@@ -2382,12 +2373,6 @@ class BatchedSimHip {
       S_.pp_nrew = reinterpret_cast<unsigned char*>(t_["pp_nrew"].data_ptr<uint8_t>());
       S_.pp_has_rl = reinterpret_cast<unsigned char*>(t_["pp_has_rl"].data_ptr<uint8_t>());
       T_PTR(x_s0, float);
-      S_.x_nsel = reinterpret_cast<short*>(t_["x_nsel"].data_ptr<int16_t>());
-      S_.x_adc = reinterpret_cast<unsigned char*>(t_["x_adc"].data_ptr<uint8_t>());
-      S_.x_ag = reinterpret_cast<unsigned char*>(t_["x_ag"].data_ptr<uint8_t>());
-      S_.x_mdc = reinterpret_cast<unsigned char*>(t_["x_mdc"].data_ptr<uint8_t>());
-      S_.x_mg = reinterpret_cast<unsigned char*>(t_["x_mg"].data_ptr<uint8_t>());
-      S_.x_has_rl = reinterpret_cast<unsigned char*>(t_["x_has_rl"].data_ptr<uint8_t>());
       T_PTR(lat_hist, int);
       S_.lat_count = reinterpret_cast<long long*>(t_["lat_count"].data_ptr<int64_t>());
       T_PTR(lat_sum, double);
