@@ -53,10 +53,19 @@ class GraphedSACStep:
     def _eager(self):
         self.agent.train_step(self._batch_from_idx(), compute_stats=False)
 
-    def _capture(self, warmup_iters: int):
+    def _fill_idx(self):
+        """Sample indices ON-DEVICE (a small randint kernel).  A host-side
+        randint + pageable H2D copy_ here serializes against every BLOCKING
+        stream via legacy-default-stream semantics — with the advance kernel
+        on the CU-masked (blocking) stream, each train step would wait out
+        the whole simulation window (measured: exactly one overlapped step
+        per cycle, at any island size / grid shape)."""
         n = max(1, self.replay.size)
-        self.idx.copy_(torch.randint(0, n, (self.batch_size,),
-                                     generator=self._cpu_gen))
+        torch.randint(0, n, (self.batch_size,), device=self.device,
+                      out=self.idx)
+
+    def _capture(self, warmup_iters: int):
+        self._fill_idx()
         try:
             side = torch.cuda.Stream(device=self.device)
             side.wait_stream(torch.cuda.current_stream(self.device))
@@ -77,9 +86,7 @@ class GraphedSACStep:
         return self.graph is not None
 
     def step(self):
-        n = max(1, self.replay.size)
-        self.idx.copy_(torch.randint(0, n, (self.batch_size,),
-                                     generator=self._cpu_gen))
+        self._fill_idx()
         if self.graph is not None:
             self.graph.replay()
         else:
