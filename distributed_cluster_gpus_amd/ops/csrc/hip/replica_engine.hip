@@ -2409,6 +2409,8 @@ class BatchedSimHip {
   // priorities cannot help because no slot ever frees.  Masking carves a
   // small CU island the advance never touches; kernels on ordinary streams
   // land there immediately.
+  void set_resident_cap(bool on) { resident_cap_on_ = on; }
+
   void enable_masked_stream(int n_reserved) {
     if (masked_stream_) return;
     reserved_cus_ = n_reserved;
@@ -2470,7 +2472,7 @@ class BatchedSimHip {
     // chsac overlapped path: cap the grid at RESIDENT capacity so no
     // pending workgroups head-of-line-block the train stream's dispatch
     // (the grid-stride loop covers the remaining replicas)
-    if (S_.algo == A_CHSAC && S_.serve_device) {
+    if (S_.algo == A_CHSAC && S_.serve_device && resident_cap_on_) {
       if (resident_blocks_ == 0) {
         int per_cu = 0;
         (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(
@@ -2522,6 +2524,7 @@ class BatchedSimHip {
   hipStream_t masked_stream_ = nullptr;
   int reserved_cus_ = 0;
   int resident_blocks_ = 0;
+  bool resident_cap_on_ = false;
 };
 
 #undef T_PTR
@@ -2540,6 +2543,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("enable_masked_stream", &dcg::BatchedSimHip::enable_masked_stream,
            py::arg("n_reserved_cus"))
       .def("masked_active", &dcg::BatchedSimHip::masked_active)
+      .def("set_resident_cap", &dcg::BatchedSimHip::set_resident_cap,
+           py::arg("on"))
       .def("advance_done", &dcg::BatchedSimHip::advance_done)
       .def("advance_sync", &dcg::BatchedSimHip::advance_sync);
   m.def("rl_forward_mfma", &dcg::rl_forward_mfma,
