@@ -33,8 +33,6 @@ class GraphedSACStep:
             "graphed step does not support the DP all-reduce hook"
         self.idx = torch.zeros(self.batch_size, dtype=torch.long,
                                device=self.device)
-        self._cpu_gen = torch.Generator().manual_seed(
-            int(torch.initial_seed()) % (2**31))
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.capture_error: Optional[str] = None
         self._capture(warmup_iters)
