@@ -1424,8 +1424,9 @@ __device__ void rl_realloc_inline(Ctx& c, double now) {
 
 // ---------------- the advance body (one replica per call) ----------------
 template <int ALGO>
-__device__ void advance_one(const EngineDesc& S, int slot_id, int lane,
-                            double t_target, long long max_ev, char* smem) {
+__device__ __attribute__((always_inline)) void advance_one(
+    const EngineDesc& S, int slot_id, int lane,
+    double t_target, long long max_ev, char* smem) {
   Ctx c;
   c.S = &S;
   c.r = slot_id;
