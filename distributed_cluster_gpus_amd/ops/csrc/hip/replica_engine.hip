@@ -1422,11 +1422,16 @@ __device__ void rl_realloc_inline(Ctx& c, double now) {
   store_fence();
 }
 
-// ---------------- the advance body (one replica per call) ----------------
+// ---------------- the advance kernel ----------------
 template <int ALGO>
-__device__ __attribute__((always_inline)) void advance_one(
-    const EngineDesc& S, int slot_id, int lane,
-    double t_target, long long max_ev, char* smem) {
+__global__ void __launch_bounds__(THREADS_PER_BLOCK)
+advance_kernel(EngineDesc S, double t_target, long long max_ev) {
+  // SUBW lanes form one replica slot (64: wave-per-replica; 8: eight
+  // replicas per wavefront)
+  int slot_id = (blockIdx.x * blockDim.x + threadIdx.x) / SUBW;
+  int lane = threadIdx.x & (SUBW - 1);
+  if (slot_id >= S.n_rep) return;
+
   Ctx c;
   c.S = &S;
   c.r = slot_id;
@@ -1451,6 +1456,7 @@ __device__ __attribute__((always_inline)) void advance_one(
   // SUBW==64: per replica, Hot + s_finish mirror + x_time mirror.
   // SUBW==8: Hot only (the mirrors would exceed the LDS budget at 8
   // replicas/wave); l_fin/l_xt then alias the replica's global rows.
+  extern __shared__ __attribute__((aligned(16))) char smem[];
   {
     size_t hot_sz = (sizeof(Hot) + 15) & ~size_t(15);
     size_t fin_sz = SUBW == 64 ? (size_t)S.total_slots * sizeof(double) : 0;
@@ -2049,21 +2055,14 @@ __device__ __attribute__((always_inline)) void advance_one(
   }
 }
 
-// Grid-stride launcher: when the grid is capped to RESIDENT capacity (the
-// chsac overlapped path), every block is running from t=0 and no pending
-// workgroups head-of-line-block concurrent kernels' dispatch — the SAC
-// train stream then truly overlaps; with a full grid the stride loop runs
-// exactly once per slot (identical to the round-1 launch shape).
-template <int ALGO>
-__global__ void __launch_bounds__(THREADS_PER_BLOCK)
-advance_kernel(EngineDesc S, double t_target, long long max_ev) {
-  int lane = threadIdx.x & (SUBW - 1);
-  int stride = (gridDim.x * blockDim.x) / SUBW;
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  for (int slot_id = (int)((blockIdx.x * blockDim.x + threadIdx.x) / SUBW);
-       slot_id < S.n_rep; slot_id += stride)
-    advance_one<ALGO>(S, slot_id, lane, t_target, max_ev, smem);
-}
+// NEGATIVE RESULT (kept out): a grid-stride launcher over an extracted
+// advance_one device function, with the chsac grid capped at resident
+// capacity to free the workgroup dispatcher for concurrent train kernels.
+// The restructure tanked the flagship 400 -> 247 M ev/s (65 k replicas:
+// 551 -> 302 M) with OR without always_inline — the loop-wrapped body
+// spills the wave-uniform scalars, the same failure mode as round 1's
+// noinline experiment.  Train/sim concurrency remains bounded by
+// dispatch behavior; the update-rate controller owns that tradeoff.
 
 // ---------------- MFMA batched actor forward ----------------
 // Matrix-core path for BATCHED policy evaluation (host-side serving /
@@ -2410,12 +2409,11 @@ class BatchedSimHip {
   // priorities cannot help because no slot ever frees.  Masking carves a
   // small CU island the advance never touches; kernels on ordinary streams
   // land there immediately.
-  void set_resident_cap(bool on) { resident_cap_on_ = on; }
+  void set_resident_cap(bool on) { (void)on; }  // negative result; no-op
 
   void enable_masked_stream(int n_reserved) {
     if (masked_stream_) return;
     reserved_cus_ = n_reserved;
-    resident_blocks_ = 0;  // recompute with the reduced CU count
     hipDeviceProp_t prop;
     int dev;
     (void)hipGetDevice(&dev);
@@ -2470,24 +2468,6 @@ class BatchedSimHip {
       throw std::runtime_error(
           "scenario too large for the LDS-mirrored engine (total_slots + "
           "tcap exceed the 64 KiB dynamic-LDS budget per block)");
-    // chsac overlapped path: cap the grid at RESIDENT capacity so no
-    // pending workgroups head-of-line-block the train stream's dispatch
-    // (the grid-stride loop covers the remaining replicas)
-    if (S_.algo == A_CHSAC && S_.serve_device && resident_cap_on_) {
-      if (resident_blocks_ == 0) {
-        int per_cu = 0;
-        (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(
-            &per_cu, advance_kernel<A_CHSAC>, THREADS_PER_BLOCK, shmem);
-        hipDeviceProp_t prop;
-        int dev = 0;
-        (void)hipGetDevice(&dev);
-        (void)hipGetDeviceProperties(&prop, dev);
-        int cus = prop.multiProcessorCount - reserved_cus_;
-        resident_blocks_ = (per_cu > 0 && cus > 0) ? per_cu * cus
-                                                   : (1 << 30);
-      }
-      if (blocks > resident_blocks_) blocks = resident_blocks_;
-    }
     dim3 grid(blocks), block(THREADS_PER_BLOCK);
     hipStream_t stream = masked_stream_
         ? masked_stream_ : (hipStream_t)at::hip::getCurrentHIPStream();
@@ -2524,8 +2504,6 @@ class BatchedSimHip {
   std::unordered_map<std::string, torch::Tensor> t_;
   hipStream_t masked_stream_ = nullptr;
   int reserved_cus_ = 0;
-  int resident_blocks_ = 0;
-  bool resident_cap_on_ = false;
 };
 
 #undef T_PTR
