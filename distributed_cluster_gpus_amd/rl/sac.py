@@ -101,43 +101,35 @@ class MaskedHybridSAC(nn.Module):
         q_min = q_min - self.alpha.detach().to(self.device).view(1,) * logp_n.view(-1, 1)
         return q_min, logp_n
 
-    def update(self, batch: Dict[str, torch.Tensor], gamma: float = 0.99,
-               tau: float = 0.005, target_entropy: float = -3.0,
-               compute_stats: bool = True):
-        dev = self.device
-        s = batch["s"].to(dev)
-        s_next = batch["s_next"].to(dev)
-        a_dc = batch["a_dc"].to(dev)
-        a_g = batch["a_g"].to(dev)
-        r_eff = batch["r_eff"].to(dev).unsqueeze(-1)
-        done = batch["done"].to(dev).unsqueeze(-1)
+    def _optimize(self, opt, loss, sync_params, clip_params=None):
+        """backward -> (optional DP grad all-reduce over sync_params) ->
+        clip(5.0) over clip_params (defaults to sync_params; the temperature
+        scalar is synced but never clipped) -> step."""
+        opt.zero_grad(set_to_none=True)
+        loss.backward()
+        if self.allreduce_hook is not None:
+            self.allreduce_hook(sync_params)
+        nn.utils.clip_grad_norm_(clip_params or sync_params, 5.0)
+        opt.step()
 
-        def _m(key, default=None):
-            m = batch.get(key, default)
-            return m.to(dev).bool() if m is not None else None
-        mask_dc = _m("mask_dc")
-        mask_g = _m("mask_g")
-        mask_dc_n = _m("mask_dc_n", batch.get("mask_dc"))
-        mask_g_n = _m("mask_g_n", batch.get("mask_g"))
-
-        # --- critic ---
+    def _critic_phase(self, s, s_next, a_dc, a_g, r_eff, done, gamma,
+                      mask_dc_n, mask_g_n):
+        """QR-Huber TD loss against entropy-regularized target quantiles."""
         h = self.encoder(s)
         h_next = self.encoder(s_next).detach()
         with torch.no_grad():
             q_next, _ = self._target_quantiles(h_next, mask_dc_n, mask_g_n)
             target = (r_eff + (1 - done) * gamma * q_next).detach()
         q1, q2 = self.critic(h, a_dc, a_g)
-        taus = self.taus.to(dev)
-        loss_critic = quantile_huber_loss(q1, target, taus) + \
+        taus = self.taus.to(self.device)
+        loss = quantile_huber_loss(q1, target, taus) + \
             quantile_huber_loss(q2, target, taus)
-        self.critic_opt.zero_grad(set_to_none=True)
-        loss_critic.backward()
-        if self.allreduce_hook is not None:
-            self.allreduce_hook(list(self.critic.parameters()))
-        nn.utils.clip_grad_norm_(self.critic.parameters(), 5.0)
-        self.critic_opt.step()
+        self._optimize(self.critic_opt, loss, list(self.critic.parameters()))
+        return loss
 
-        # --- actor & temperature ---
+    def _actor_phase(self, s, mask_dc, mask_g, target_entropy):
+        """-Q + alpha*logp actor loss joint with the learned-temperature
+        objective (one backward over both)."""
         h_pi = self.encoder(s)
         logits_dc, logits_g = self.actor(h_pi)
         a_dc_pi, logp_dc = sample_categorical(logits_dc, mask_dc)
@@ -147,19 +139,37 @@ class MaskedHybridSAC(nn.Module):
         q_pi = torch.min(q1_pi, q2_pi).mean(dim=-1)
         actor_loss = (self.alpha.detach() * logp - q_pi).mean()
         temp_loss = -(self.log_alpha * (logp.detach() + target_entropy)).mean()
-        self.actor_opt.zero_grad(set_to_none=True)
-        (actor_loss + temp_loss).backward()
-        if self.allreduce_hook is not None:
-            self.allreduce_hook(list(self.encoder.parameters()) +
-                                list(self.actor.parameters()) + [self.log_alpha])
-        nn.utils.clip_grad_norm_(
-            list(self.encoder.parameters()) + list(self.actor.parameters()), 5.0)
-        self.actor_opt.step()
+        shared = list(self.encoder.parameters()) + \
+            list(self.actor.parameters())
+        self._optimize(self.actor_opt, actor_loss + temp_loss,
+                       shared + [self.log_alpha], clip_params=shared)
+        return actor_loss, temp_loss
 
-        # --- polyak target update ---
-        with torch.no_grad():
-            for tp, p in zip(self.target_critic.parameters(), self.critic.parameters()):
-                tp.data.mul_(1 - tau).add_(tau * p.data)
+    @torch.no_grad()
+    def _polyak(self, tau):
+        for tp, p in zip(self.target_critic.parameters(),
+                         self.critic.parameters()):
+            tp.data.mul_(1 - tau).add_(tau * p.data)
+
+    def update(self, batch: Dict[str, torch.Tensor], gamma: float = 0.99,
+               tau: float = 0.005, target_entropy: float = -3.0,
+               compute_stats: bool = True):
+        dev = self.device
+        s = batch["s"].to(dev)
+
+        def _m(key, default=None):
+            m = batch.get(key, default)
+            return m.to(dev).bool() if m is not None else None
+
+        loss_critic = self._critic_phase(
+            s, batch["s_next"].to(dev), batch["a_dc"].to(dev),
+            batch["a_g"].to(dev), batch["r_eff"].to(dev).unsqueeze(-1),
+            batch["done"].to(dev).unsqueeze(-1), gamma,
+            _m("mask_dc_n", batch.get("mask_dc")),
+            _m("mask_g_n", batch.get("mask_g")))
+        actor_loss, temp_loss = self._actor_phase(
+            s, _m("mask_dc"), _m("mask_g"), target_entropy)
+        self._polyak(tau)
 
         if not compute_stats:
             return {}
