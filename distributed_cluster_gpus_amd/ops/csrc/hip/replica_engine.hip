@@ -717,7 +717,12 @@ __device__ void emit_job_row(Ctx& c, int d, int jt, int jid, int ing,
 //   and re-estimates total power before the next deficit check.
 // Task order = DC-major, then running_jobs insertion order (s_seq); step
 // order = i0-k ascending — exactly the reference's stable rho sort.
-__device__ void cap_greedy_control(Ctx& c, double now) {
+__device__ __attribute__((noinline)) void cap_greedy_control(Ctx& c,
+                                                             double now) {
+  // noinline ON PURPOSE: this runs once per log tick (cold), but inlined
+  // it inflates live ranges across the whole advance loop — the
+  // cap_greedy kernel instantiation measured 233 M ev/s vs 426 M for the
+  // identical trajectory under cap_uniform (round2_algo_sweep.jsonl).
   const EngineDesc& S = *c.S;
   // sorted ladder (atoms_for_task sorts freq_levels)
   double lv[MAX_FREQ];
