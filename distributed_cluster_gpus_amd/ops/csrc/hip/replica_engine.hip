@@ -642,7 +642,8 @@ __device__ void accrue_to(Ctx& c, double t) {
 }
 
 // emit one cluster-log row set (logging replica only; wave-cooperative counts)
-__device__ void emit_cluster_rows(Ctx& c, double now) {
+__device__ __attribute__((noinline)) void emit_cluster_rows(Ctx& c,
+                                                             double now) {
   const EngineDesc& S = *c.S;
   if (c.r != S.log_replica) return;
   int64_t base = (int64_t)c.r * S.total_slots;
@@ -1207,7 +1208,8 @@ __device__ void rl_emit_transition(Ctx& c, const float* s0, int a_dc, int a_g,
 // replica's pool, then reallocate them one by one with fresh RL actions
 // (reference _preempt_all_training_jobs :396-409 + _rl_reallocate_training_jobs
 // :498-534; a failed resume re-queues instead of stranding — oracle fix) ----
-__device__ int rl_elastic_preempt_all(Ctx& c, int d, double now) {
+__device__ __attribute__((noinline)) int rl_elastic_preempt_all(
+    Ctx& c, int d, double now) {
   const EngineDesc& S = *c.S;
   int64_t base = (int64_t)c.r * S.total_slots;
   int lo = S.slot_off[d], hi = S.slot_off[d + 1];
@@ -1399,7 +1401,8 @@ __device__ void rl_drain_inline(Ctx& c, int d, double now) {
 // in-kernel elastic reallocation chain (serve_device): fresh obs + actor
 // forward per pool entry, all within this launch (reference
 // _rl_reallocate_training_jobs :498-534; failed resumes re-queue)
-__device__ void rl_realloc_inline(Ctx& c, double now) {
+__device__ __attribute__((noinline)) void rl_realloc_inline(Ctx& c,
+                                                            double now) {
   const EngineDesc& S = *c.S;
   int cnt = S.pp_count[c.r];
   for (int cur = 0; cur < cnt; ++cur) {
