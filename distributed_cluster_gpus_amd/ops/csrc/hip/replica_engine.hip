@@ -642,8 +642,7 @@ __device__ void accrue_to(Ctx& c, double t) {
 }
 
 // emit one cluster-log row set (logging replica only; wave-cooperative counts)
-__device__ __attribute__((noinline)) void emit_cluster_rows(Ctx& c,
-                                                             double now) {
+__device__ void emit_cluster_rows(Ctx& c, double now) {
   const EngineDesc& S = *c.S;
   if (c.r != S.log_replica) return;
   int64_t base = (int64_t)c.r * S.total_slots;
