@@ -104,3 +104,23 @@ def test_plot_sh_wrapper(two_runs, tmp_path):
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-1500:]
     assert os.path.exists(os.path.join(root, "report", "summary.csv"))
+
+
+def test_violin_and_boxen_render(tmp_path):
+    """The violin / letter-value latency views render as PNGs (matplotlib is
+    present here) and always emit their CSV data artifacts."""
+    import pandas as pd
+    from distributed_cluster_gpus_amd.analysis.render import emit, have_mpl
+    rng = np.random.default_rng(0)
+    df = pd.DataFrame({
+        "latency_s": np.concatenate([rng.exponential(0.01, 400),
+                                     rng.exponential(0.05, 400)]),
+        "algo": ["a"] * 400 + ["b"] * 400,
+    })
+    for kind in ("violin", "boxen"):
+        path = emit(df, str(tmp_path), f"lat_{kind}", kind=kind,
+                    y="latency_s", hue="algo", title=kind)
+        assert os.path.exists(os.path.join(str(tmp_path), f"lat_{kind}.csv"))
+        if have_mpl():
+            assert os.path.exists(os.path.join(str(tmp_path),
+                                               f"lat_{kind}.png"))
