@@ -109,6 +109,7 @@ def test_plot_sh_wrapper(two_runs, tmp_path):
 def test_violin_and_boxen_render(tmp_path):
     """The violin / letter-value latency views render as PNGs (matplotlib is
     present here) and always emit their CSV data artifacts."""
+    import numpy as np
     import pandas as pd
     from distributed_cluster_gpus_amd.analysis.render import emit, have_mpl
     rng = np.random.default_rng(0)
